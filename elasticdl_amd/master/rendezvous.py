@@ -1,0 +1,155 @@
+"""Elastic rendezvous for RCCL process groups.
+
+Capability equivalent of the reference's HorovodRendezvousServer
+(elasticdl/python/master/rendezvous_server.py:34-167), rebuilt for
+torch.distributed over RCCL: the master hosts one long-lived
+``torch.distributed.TCPStore``; each rendezvous generation is a key
+prefix ``"<rendezvous_id>/"`` inside that store, so re-forming the world
+after elasticity events never reuses stale bootstrap keys.
+
+State machine (mirrors the reference):
+
+- ``add_worker`` / ``remove_worker`` stage the *next* host list;
+- ``get_worker_host_rank`` flips to the staged world once every member of
+  the current world has reported ready (polled its rank), bumping the
+  monotonically increasing ``rendezvous_id``;
+- a worker not in the current world receives rank -1 and keeps polling.
+
+Workers consume this through
+elasticdl_amd.collective.communicator.CommunicatorManager, which tears
+down and rebuilds the RCCL communicator whenever ``rendezvous_id``
+changes — the RCCL analog of the reference's hvd.shutdown()/init() cycle.
+"""
+
+import copy
+import threading
+import time
+from datetime import timedelta
+from typing import List, Optional
+
+from elasticdl_amd.common.log_utils import default_logger as logger
+
+
+class ElasticRendezvousServer:
+    def __init__(self, host: str, port: int = 0):
+        self._host = host
+        self._requested_port = port
+        self._port: Optional[int] = None
+        self._store = None
+        self._lock = threading.Lock()
+        self._rendezvous_id = 0
+        self._cur_hosts: List[str] = []
+        self._next_hosts: Optional[List[str]] = None
+        self._ready_hosts = set()
+        self._dead_hosts = set()
+        self._cur_completed = True
+        # flip debounce: after staging, wait a moment so racing removals of
+        # the same elasticity event coalesce into one generation
+        self._flip_delay_sec = 0.5
+
+    # ------------------------------------------------------------- lifecycle
+    def start(self) -> int:
+        from torch.distributed import TCPStore
+
+        port = self._requested_port or _free_port()
+        self._store = TCPStore(
+            self._host if self._host not in ("", "0.0.0.0") else "127.0.0.1",
+            port,
+            is_master=True,
+            timeout=timedelta(seconds=3600),
+        )
+        self._port = port
+        logger.info("Rendezvous TCPStore listening on %s:%d", self._host, port)
+        return port
+
+    @property
+    def port(self) -> int:
+        return self._port
+
+    @property
+    def rendezvous_id(self) -> int:
+        return self._rendezvous_id
+
+    def world_size(self) -> int:
+        return len(self._cur_hosts)
+
+    # ------------------------------------------------------------ membership
+    def add_worker(self, worker_host: str) -> None:
+        with self._lock:
+            if not worker_host:
+                return
+            if self._next_hosts is None:
+                self._next_hosts = copy.deepcopy(self._cur_hosts)
+            # after training started, never resurrect an empty world
+            # (reference: rendezvous_server.py add_worker)
+            if self._rendezvous_id > 0 and not self._next_hosts:
+                return
+            if worker_host not in self._next_hosts:
+                self._next_hosts.append(worker_host)
+                logger.info(
+                    "Rendezvous: staged add of %s (next world %s)",
+                    worker_host,
+                    self._next_hosts,
+                )
+
+    def remove_worker(self, worker_host: str) -> None:
+        with self._lock:
+            if worker_host in self._cur_hosts or (
+                self._next_hosts and worker_host in self._next_hosts
+            ):
+                if self._next_hosts is None:
+                    self._next_hosts = copy.deepcopy(self._cur_hosts)
+                if worker_host in self._next_hosts:
+                    self._next_hosts.remove(worker_host)
+                # a dead member can never report ready — drop it from the
+                # current world's readiness requirement so the flip is not
+                # deadlocked (improves on the reference, which can stall if
+                # a worker dies between world formation and completion)
+                self._dead_hosts.add(worker_host)
+                logger.info(
+                    "Rendezvous: staged removal of %s (next world %s)",
+                    worker_host,
+                    self._next_hosts,
+                )
+
+    # ----------------------------------------------------------------- query
+    def get_comm_rank(self, worker_host: str) -> dict:
+        """One-stop poll for workers: rank/world/rendezvous_id/store addr."""
+        with self._lock:
+            if self._next_hosts is not None and self._cur_completed:
+                time.sleep(self._flip_delay_sec)
+                self._flip()
+            if worker_host not in self._cur_hosts:
+                rank = -1
+            else:
+                if not self._cur_completed:
+                    self._ready_hosts.add(worker_host)
+                    if self._ready_hosts >= set(self._cur_hosts) - self._dead_hosts:
+                        self._cur_completed = True
+                        self._ready_hosts = set()
+                rank = self._cur_hosts.index(worker_host)
+            return {
+                "rank_id": rank,
+                "world_size": len(self._cur_hosts),
+                "rendezvous_id": self._rendezvous_id,
+                "rendezvous_port": self._port or 0,
+            }
+
+    def _flip(self) -> None:
+        self._cur_hosts = self._next_hosts
+        self._next_hosts = None
+        self._rendezvous_id += 1
+        self._cur_completed = False
+        self._ready_hosts = set()
+        self._dead_hosts = set()
+        logger.info(
+            "Rendezvous %d: world=%s", self._rendezvous_id, self._cur_hosts
+        )
+
+
+def _free_port() -> int:
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]

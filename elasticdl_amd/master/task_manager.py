@@ -1,0 +1,372 @@
+"""Dynamic data-sharding task dispatcher.
+
+Re-implements the behavior of the reference's TaskManager
+(elasticdl/python/master/task_manager.py:95-616):
+
+- training data is partitioned into tasks (shards of record ranges) from a
+  data reader's shard list or from (dataset_size, batch_size,
+  num_minibatches_per_task);
+- workers pull tasks (todo -> doing, stamped with worker id + time) and
+  report results; failed tasks are requeued up to MAX_TASK_RETRIES times;
+- tasks owned by a dead worker are recovered (requeued);
+- when the training todo queue drains, the next epoch is created (with
+  optional shard- and record-level shuffling); after the last epoch a
+  train-end callback task is emitted once all training tasks finished;
+- evaluation tasks can be injected at any time (version-triggered);
+- a watchdog thread re-assigns tasks whose runtime exceeds
+  max(min_timeout, 3 x slowest observed task) and notifies callbacks so
+  the hung worker can be deleted;
+- supports max-step stop and resuming the completed-step counter from a
+  checkpoint directory.
+"""
+
+import random
+import threading
+import time
+from typing import Callable, Dict, List, Optional, Tuple
+
+from elasticdl_amd.common.constants import MAX_TASK_RETRIES
+from elasticdl_amd.common.log_utils import default_logger as logger
+from elasticdl_amd.common.task import Shard, Task, TaskType
+
+_MIN_TASK_TIMEOUT_SEC = 300.0
+
+
+class _DoingEntry:
+    __slots__ = ("task", "worker_id", "start_time")
+
+    def __init__(self, task: Task, worker_id: int, start_time: float):
+        self.task = task
+        self.worker_id = worker_id
+        self.start_time = start_time
+
+
+class TaskManager:
+    def __init__(
+        self,
+        training_shards: Optional[List[Tuple[str, int, int]]] = None,
+        evaluation_shards: Optional[List[Tuple[str, int, int]]] = None,
+        prediction_shards: Optional[List[Tuple[str, int, int]]] = None,
+        records_per_task: int = 0,
+        num_epochs: int = 1,
+        max_step: int = 0,
+        shuffle: bool = False,
+        shuffle_shards: bool = False,
+        task_timeout_sec: float = _MIN_TASK_TIMEOUT_SEC,
+    ):
+        self._lock = threading.Lock()
+        self._training_shards = list(training_shards or [])
+        self._evaluation_shards = list(evaluation_shards or [])
+        self._prediction_shards = list(prediction_shards or [])
+        self._records_per_task = records_per_task
+        self._num_epochs = num_epochs
+        self._epoch = 0
+        self._max_step = max_step
+        self._completed_steps = 0
+        self._shuffle = shuffle
+        self._shuffle_shards = shuffle_shards
+        self._task_timeout_sec = max(task_timeout_sec, 1e-3)
+
+        self._todo: List[Task] = []
+        self._doing: Dict[int, _DoingEntry] = {}
+        self._task_id = 0
+        self._task_retry_count: Dict[int, int] = {}
+        self._max_task_completed_time = 0.0
+
+        self.total_records = sum(s[2] - s[1] for s in self._training_shards)
+        self.failed_records = 0
+
+        self._train_end_callback_emitted = False
+        self._train_end_callback_done = False
+        self._eval_todo_count = 0
+        self._task_timeout_callbacks: List[Callable[[int], None]] = []
+        self._version_holder: Callable[[], int] = lambda: -1
+        self._worker_done: Dict[int, bool] = {}
+        self._stop_watchdog = threading.Event()
+        self._watchdog: Optional[threading.Thread] = None
+
+        if self._training_shards:
+            self._create_training_tasks()
+            logger.info(
+                "TaskManager: %d training tasks for epoch 0 (%d records)",
+                len(self._todo),
+                self.total_records,
+            )
+
+    # ------------------------------------------------------------------ build
+    def set_version_holder(self, fn: Callable[[], int]) -> None:
+        """Model-version supplier stamped onto evaluation tasks."""
+        self._version_holder = fn
+
+    def _next_task_id(self) -> int:
+        self._task_id += 1
+        return self._task_id
+
+    def _shards_to_tasks(
+        self, shards: List[Tuple[str, int, int]], task_type: str
+    ) -> List[Task]:
+        tasks = []
+        shards = list(shards)
+        if task_type == TaskType.TRAINING and self._shuffle_shards:
+            random.shuffle(shards)
+        for name, start, end in shards:
+            step = self._records_per_task if self._records_per_task > 0 else end - start
+            indices = None
+            if task_type == TaskType.TRAINING and self._shuffle:
+                indices = list(range(start, end))
+                random.shuffle(indices)
+            for lo in range(start, end, step):
+                hi = min(lo + step, end)
+                sub_indices = None
+                if indices is not None:
+                    sub_indices = indices[lo - start:hi - start]
+                tasks.append(
+                    Task(
+                        task_id=self._next_task_id(),
+                        shard=Shard(name, lo, hi, sub_indices),
+                        type=task_type,
+                    )
+                )
+        return tasks
+
+    def _create_training_tasks(self) -> None:
+        self._todo.extend(self._shards_to_tasks(self._training_shards, TaskType.TRAINING))
+
+    def set_training_params(
+        self,
+        dataset_size: int,
+        batch_size: int,
+        num_epochs: int = 1,
+        num_minibatches_per_shard: int = 1,
+        shuffle: bool = False,
+        shuffle_shards: bool = False,
+    ) -> None:
+        """Worker-driven shard creation (reference: task_manager.py:283-295):
+        tasks cover [0, dataset_size) in chunks of
+        batch_size * num_minibatches_per_shard records."""
+        with self._lock:
+            if self._training_shards:
+                return
+            self._training_shards = [("", 0, dataset_size)]
+            self._records_per_task = max(1, batch_size * num_minibatches_per_shard)
+            self._num_epochs = num_epochs
+            self._shuffle = shuffle
+            self._shuffle_shards = shuffle_shards
+            self.total_records = dataset_size
+            self._create_training_tasks()
+            logger.info(
+                "TaskManager: %d worker-defined training tasks (%d records)",
+                len(self._todo),
+                dataset_size,
+            )
+
+    def create_evaluation_tasks(self, model_version: int = -1) -> int:
+        """Inject evaluation tasks (reference: task_manager create_evaluation_tasks)."""
+        with self._lock:
+            tasks = self._shards_to_tasks(self._evaluation_shards, TaskType.EVALUATION)
+            for t in tasks:
+                t.model_version = model_version
+            # evaluation tasks go to the head so they interleave promptly
+            self._todo = tasks + self._todo
+            self._eval_todo_count += len(tasks)
+            return len(tasks)
+
+    def create_prediction_tasks(self) -> int:
+        with self._lock:
+            tasks = self._shards_to_tasks(self._prediction_shards, TaskType.PREDICTION)
+            self._todo.extend(tasks)
+            return len(tasks)
+
+    def create_train_end_callback_task(self) -> None:
+        """One train-end task, dispatched to a single worker for final export
+        (reference: task_manager.py:394-429)."""
+        self._todo.append(
+            Task(
+                task_id=self._next_task_id(),
+                shard=None,
+                type=TaskType.TRAIN_END_CALLBACK,
+                model_version=self._version_holder(),
+            )
+        )
+        self._train_end_callback_emitted = True
+
+    # ---------------------------------------------------------------- workers
+    def register_task_timeout_callback(self, fn: Callable[[int], None]) -> None:
+        self._task_timeout_callbacks.append(fn)
+
+    def get(self, worker_id: int) -> Task:
+        """Pop the next task for a worker; WAIT when work is pending
+        elsewhere; NONE when the job has no further work."""
+        with self._lock:
+            if self._max_step and self._completed_steps >= self._max_step:
+                return Task(task_id=0, shard=None, type=TaskType.NONE)
+            if not self._todo:
+                if self._maybe_start_next_epoch():
+                    pass  # fall through with refilled todo
+                elif self._should_emit_train_end_callback():
+                    self.create_train_end_callback_task()
+                elif self._doing or (
+                    self._train_end_callback_emitted
+                    and not self._train_end_callback_done
+                ):
+                    return Task(task_id=0, shard=None, type=TaskType.WAIT)
+                else:
+                    return Task(task_id=0, shard=None, type=TaskType.NONE)
+            task = self._todo.pop(0)
+            if task.type == TaskType.EVALUATION and task.model_version < 0:
+                task.model_version = self._version_holder()
+            self._doing[task.task_id] = _DoingEntry(task, worker_id, time.monotonic())
+            return task
+
+    def _maybe_start_next_epoch(self) -> bool:
+        if not self._training_shards:
+            return False
+        # only start the next epoch once all in-flight training tasks resolve?
+        # The reference refills as soon as todo drains (task_manager.py:453-459).
+        if self._epoch + 1 >= self._num_epochs:
+            return False
+        self._epoch += 1
+        self._create_training_tasks()
+        logger.info("TaskManager: starting epoch %d", self._epoch)
+        return bool(self._todo)
+
+    def _training_finished(self) -> bool:
+        no_training_left = not any(
+            t.type == TaskType.TRAINING for t in self._todo
+        ) and not any(
+            e.task.type == TaskType.TRAINING for e in self._doing.values()
+        )
+        return no_training_left and self._epoch + 1 >= self._num_epochs
+
+    def _should_emit_train_end_callback(self) -> bool:
+        return (
+            self._train_end_callback_enabled
+            and not self._train_end_callback_emitted
+            and self._training_shards
+            and self._training_finished()
+        )
+
+    # train-end callback emission is opt-in (set by the job service when the
+    # model spec has export callbacks)
+    _train_end_callback_enabled = False
+
+    def enable_train_end_callback(self) -> None:
+        self._train_end_callback_enabled = True
+
+    def report(self, task_id: int, success: bool, worker_id: int = -1) -> Tuple[bool, Task]:
+        """Report task completion. Returns (was_in_doing, task)."""
+        with self._lock:
+            entry = self._doing.pop(task_id, None)
+            if entry is None:
+                return False, None
+            task = entry.task
+            elapsed = time.monotonic() - entry.start_time
+            if success:
+                self._max_task_completed_time = max(
+                    self._max_task_completed_time, elapsed
+                )
+                if task.type == TaskType.TRAINING:
+                    self._completed_steps += 1
+                elif task.type == TaskType.EVALUATION:
+                    self._eval_todo_count = max(0, self._eval_todo_count - 1)
+                elif task.type == TaskType.TRAIN_END_CALLBACK:
+                    self._train_end_callback_done = True
+            else:
+                retries = self._task_retry_count.get(task_id, 0) + 1
+                self._task_retry_count[task_id] = retries
+                if retries <= MAX_TASK_RETRIES:
+                    logger.info(
+                        "Task %d failed (retry %d/%d); requeueing",
+                        task_id,
+                        retries,
+                        MAX_TASK_RETRIES,
+                    )
+                    self._todo.append(task)
+                else:
+                    logger.error("Task %d exceeded max retries; dropping", task_id)
+                    if task.shard is not None:
+                        self.failed_records += task.shard.size
+                    if task.type == TaskType.TRAIN_END_CALLBACK:
+                        self._train_end_callback_done = True
+            return True, task
+
+    def recover_tasks(self, worker_id: int) -> int:
+        """Requeue all doing tasks of a dead worker
+        (reference: task_manager.py:544-560)."""
+        with self._lock:
+            ids = [
+                tid
+                for tid, e in self._doing.items()
+                if e.worker_id == worker_id
+            ]
+            for tid in ids:
+                entry = self._doing.pop(tid)
+                self._todo.append(entry.task)
+            if ids:
+                logger.info(
+                    "Recovered %d tasks from worker %d", len(ids), worker_id
+                )
+            return len(ids)
+
+    # --------------------------------------------------------------- watchdog
+    def start(self) -> None:
+        if self._watchdog is None:
+            self._watchdog = threading.Thread(
+                target=self._watchdog_loop, name="task-watchdog", daemon=True
+            )
+            self._watchdog.start()
+
+    def stop(self) -> None:
+        self._stop_watchdog.set()
+
+    def _watchdog_loop(self) -> None:
+        while not self._stop_watchdog.wait(min(30.0, self._task_timeout_sec / 3)):
+            self._reassign_timeout_tasks()
+
+    def _reassign_timeout_tasks(self) -> None:
+        """Reference: task_manager.py:592-616 — requeue tasks running longer
+        than max(timeout, 3 x slowest completed task) and report the worker."""
+        threshold = max(self._task_timeout_sec, 3 * self._max_task_completed_time)
+        now = time.monotonic()
+        victims: List[int] = []
+        with self._lock:
+            for tid, e in list(self._doing.items()):
+                if now - e.start_time > threshold:
+                    victims.append(e.worker_id)
+                    self._todo.append(e.task)
+                    del self._doing[tid]
+        for worker_id in set(victims):
+            logger.warning("Task timeout: worker %d considered hung", worker_id)
+            for cb in self._task_timeout_callbacks:
+                cb(worker_id)
+
+    # ----------------------------------------------------------------- status
+    def finished(self) -> bool:
+        with self._lock:
+            if self._max_step and self._completed_steps >= self._max_step:
+                return True
+            if self._todo or self._doing:
+                return False
+            if self._training_shards and self._epoch + 1 < self._num_epochs:
+                return False
+            if self._train_end_callback_enabled and self._training_shards:
+                return self._train_end_callback_done
+            return True
+
+    @property
+    def completed_steps(self) -> int:
+        return self._completed_steps
+
+    def set_completed_steps(self, steps: int) -> None:
+        """Resume from checkpoint (reference: task_manager.py:208-221)."""
+        self._completed_steps = steps
+
+    def counts(self) -> Dict[str, int]:
+        with self._lock:
+            return {
+                "todo": len(self._todo),
+                "doing": len(self._doing),
+                "completed_steps": self._completed_steps,
+                "epoch": self._epoch,
+                "failed_records": self.failed_records,
+            }
