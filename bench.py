@@ -1,0 +1,256 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet50 synthetic-ImageNet AllReduce DP (bf16).
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches via torch.distributed.run (one rank per GPU,
+RCCL over xGMI); rank 0 prints ONE JSON line with the whole-job
+samples/sec. Also supports the PS-path workloads (--model deepfm|wide_deep)
+with a colocated GPU parameter-server engine.
+
+Baseline anchor: reference ResNet50 ImageNet bs=64 on
+1xP100 = 145 images/s (docs/benchmark/ftlib_benchmark.md:119-123).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def get_dist_env():
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    return rank, world, local_rank
+
+
+def setup_dist(world):
+    import torch.distributed as dist
+
+    if world > 1 and not dist.is_initialized():
+        dist.init_process_group("nccl" if torch.cuda.is_available() else "gloo")
+    return dist
+
+
+def sync_all(dist, world, device):
+    if torch.cuda.is_available():
+        torch.cuda.synchronize(device)
+    if world > 1:
+        dist.barrier()
+        torch.cuda.synchronize(device)
+
+
+def max_over_ranks(dist, world, value: float, device) -> float:
+    if world <= 1:
+        return value
+    t = torch.tensor([value], dtype=torch.float64,
+                     device=device if torch.cuda.is_available() else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
+
+
+def bench_resnet50(args, rank, world, local_rank):
+    from elasticdl_amd.collective.distributed_optimizer import DistributedOptimizer
+    from elasticdl_amd.models import resnet
+
+    dist = setup_dist(world)
+    device = torch.device("cuda", local_rank) if torch.cuda.is_available() else torch.device("cpu")
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
+
+    torch.manual_seed(1234)
+    model = resnet.resnet50(num_classes=args.num_classes)
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    model = model.to(device, dtype).to(memory_format=torch.channels_last)
+
+    opt = DistributedOptimizer(
+        model, lr=0.1, momentum=0.9, bucket_cap_mb=args.bucket_mb
+    )
+    if world > 1:
+        for b in opt.buckets:  # one broadcast per flat buffer, rank-0 state
+            dist.broadcast(b.param_flat, 0)
+            b.master.copy_(b.param_flat.float())
+
+    bs = args.batch_size
+    n_batches = 4  # rotate a few resident synthetic batches
+    gen = torch.Generator().manual_seed(42 + rank)
+    images = [
+        torch.randn(bs, 3, args.image_size, args.image_size, generator=gen)
+        .to(device, dtype)
+        .contiguous(memory_format=torch.channels_last)
+        for _ in range(n_batches)
+    ]
+    labels = [
+        torch.randint(0, args.num_classes, (bs,), generator=gen).to(device)
+        for _ in range(n_batches)
+    ]
+
+    def one_step(i):
+        x, y = images[i % n_batches], labels[i % n_batches]
+        opt.zero_grad()
+        out = model(x)
+        loss = torch.nn.functional.cross_entropy(out.float(), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for i in range(args.warmup):
+        one_step(i)
+    sync_all(dist, world, device)
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    sync_all(dist, world, device)
+    elapsed = time.perf_counter() - t0
+    elapsed = max_over_ranks(dist, world, elapsed, device)
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1e3
+        value = bs * world * args.steps / elapsed  # whole-job images/sec
+        print(json.dumps({
+            "metric": "samples_per_sec",
+            "value": round(value, 2),
+            "unit": "images/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / 145.0, 2),
+            "dtype": "bf16" if device.type == "cuda" else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "resnet50",
+                "global_batch": bs * world,
+                "image_size": args.image_size,
+                "parallelism": f"dp{world}",
+            },
+        }))
+
+
+def bench_ps_model(args, rank, world, local_rank):
+    """PS path: DeepFM/Wide&Deep with a colocated GPU PS engine per rank
+    (embedding shard in local HBM; dense towers via the MFMA GEMM)."""
+    from elasticdl_amd.common.tensor_utils import merge_indexed_slices
+    from elasticdl_amd.layers.embedding import bind_local_engine, find_edl_embeddings
+    from elasticdl_amd.models import deepfm, wide_deep
+    from elasticdl_amd.ps.engine import PSEngine
+
+    dist = setup_dist(world)
+    device = torch.device("cuda", local_rank) if torch.cuda.is_available() else torch.device("cpu")
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
+    torch.manual_seed(1234 + rank)
+
+    zoo = deepfm if args.model == "deepfm" else wide_deep
+    if args.model == "deepfm":
+        model = deepfm.DeepFM(max_rows=args.table_rows)
+        batch_fn = lambda s: deepfm.synthetic_batch(args.batch_size, seed=s)
+    else:
+        model = wide_deep.WideDeep(max_rows=args.table_rows)
+        batch_fn = lambda s: wide_deep.synthetic_batch(args.batch_size, seed=s)
+    model = model.to(device)
+
+    opt_type, opt_args = zoo.optimizer()
+    engine = PSEngine(
+        opt_type=opt_type, opt_args=opt_args,
+        device=device, use_async=True,
+        embedding_max_rows=args.table_rows,
+    )
+    engine.push_model(
+        {},
+        [e.table_info() for e in find_edl_embeddings(model)],
+    )
+    bind_local_engine(model, engine)
+    sink = []
+    for e in find_edl_embeddings(model):
+        e.set_grad_sink(sink)
+
+    dense_opt = torch.optim.Adam(
+        [p for p in model.parameters() if p.requires_grad], lr=1e-3
+    )
+
+    batches = [batch_fn(s) for s in range(4)]
+    batches = [(ids.to(device), y.to(device)) for ids, y in batches]
+
+    def one_step(i):
+        ids, y = batches[i % len(batches)]
+        sink.clear()
+        dense_opt.zero_grad(set_to_none=True)
+        out = model(ids)
+        loss = zoo.loss(out, y)
+        loss.backward()
+        dense_opt.step()
+        by_name = {}
+        for n, s in sink:
+            by_name.setdefault(n, []).append(s)
+        merged = {n: merge_indexed_slices(*lst) for n, lst in by_name.items()}
+        engine.push_gradients({}, merged, version=0)
+        return loss
+
+    for i in range(args.warmup):
+        one_step(i)
+    sync_all(dist, world, device)
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    sync_all(dist, world, device)
+    elapsed = time.perf_counter() - t0
+    elapsed = max_over_ranks(dist, world, elapsed, device)
+
+    if rank == 0:
+        value = args.batch_size * world * args.steps / elapsed
+        print(json.dumps({
+            "metric": "samples_per_sec",
+            "value": round(value, 2),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if device.type == "cuda" else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch_size * world,
+                "table_rows": args.table_rows,
+                "parallelism": f"dp{world}+ps",
+            },
+        }))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--model", default="resnet50",
+                    choices=["resnet50", "deepfm", "wide_deep"])
+    ap.add_argument("--batch-size", "--batch_size", type=int, default=None)
+    ap.add_argument("--image-size", "--image_size", type=int, default=224)
+    ap.add_argument("--num-classes", "--num_classes", type=int, default=1000)
+    ap.add_argument("--bucket-mb", "--bucket_mb", type=float, default=25.0)
+    ap.add_argument("--table-rows", "--table_rows", type=int, default=1 << 22)
+    args = ap.parse_args()
+
+    if args.batch_size is None:
+        args.batch_size = 256 if args.model == "resnet50" else 4096
+        if not torch.cuda.is_available():
+            args.batch_size = 16
+
+    rank, world, local_rank = get_dist_env()
+    if args.model == "resnet50":
+        bench_resnet50(args, rank, world, local_rank)
+    else:
+        bench_ps_model(args, rank, world, local_rank)
+
+
+if __name__ == "__main__":
+    main()

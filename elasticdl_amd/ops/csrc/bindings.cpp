@@ -1,0 +1,293 @@
+// Python bindings for the MI355X PS kernels (ps_kernels.hip) and GEMM
+// (gemm_bf16.hip). Thin argument-checking layer: all compute stays in the
+// HIP translation units; this file never includes HIP headers so it
+// compiles as plain C++ against libtorch.
+
+#include <torch/extension.h>
+
+#include <ATen/cuda/CUDAContext.h>
+
+#include <cstdint>
+
+// hipStream_t without pulling in hip_runtime.h
+struct ihipStream_t;
+typedef struct ihipStream_t* hipStream_t;
+
+extern "C" {
+void edl_dense_sgd(float*, const float*, int64_t, float, hipStream_t);
+void edl_dense_momentum(float*, float*, const float*, int64_t, float, float,
+                        bool, hipStream_t);
+void edl_dense_adam(float*, float*, float*, float*, const float*, int64_t,
+                    float, float, float, float, hipStream_t);
+void edl_dense_adagrad(float*, float*, const float*, int64_t, float, float,
+                       hipStream_t);
+void edl_dense_ftrl(float*, float*, float*, const float*, int64_t, float,
+                    float, float, float, hipStream_t);
+void edl_sparse_sgd(float*, const float*, const int32_t*, int64_t, int64_t,
+                    float, hipStream_t);
+void edl_sparse_momentum(float*, float*, const float*, const int32_t*,
+                         int64_t, int64_t, float, float, bool, hipStream_t);
+void edl_sparse_adam(float*, float*, float*, float*, const float*,
+                     const int32_t*, int64_t, int64_t, float, float, float,
+                     float, hipStream_t);
+void edl_sparse_adagrad(float*, float*, const float*, const int32_t*, int64_t,
+                        int64_t, float, float, hipStream_t);
+void edl_sparse_ftrl(float*, float*, float*, const float*, const int32_t*,
+                     int64_t, int64_t, float, float, float, float,
+                     hipStream_t);
+void edl_ht_lookup_or_insert(int64_t*, int32_t*, int64_t, int32_t*, int32_t,
+                             const int64_t*, int64_t, int32_t*, uint8_t*,
+                             int32_t*, hipStream_t);
+void edl_ht_lookup(const int64_t*, const int32_t*, int64_t, const int64_t*,
+                   int64_t, int32_t*, hipStream_t);
+void edl_init_new_rows(float*, const int32_t*, const uint8_t*, int64_t,
+                       int64_t, uint64_t, float, float, hipStream_t);
+void edl_gather_rows(const float*, const int32_t*, int64_t, int64_t, float*,
+                     hipStream_t);
+void edl_scatter_rows(float*, const int32_t*, const float*, int64_t, int64_t,
+                      hipStream_t);
+void edl_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
+                            int, int, int, int, hipStream_t);
+void edl_fused_sgd_bf16(void*, float*, float*, const void*, int64_t, float,
+                        float, bool, float, float, hipStream_t);
+void edl_fused_adamw_bf16(void*, float*, float*, float*, const void*, int64_t,
+                          float, float, float, float, float, float, float,
+                          hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return reinterpret_cast<hipStream_t>(
+      at::cuda::getCurrentCUDAStream().stream());
+}
+
+void check_f32_cuda(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be float32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// ------------------------------ dense optimizers ------------------------
+void dense_sgd(torch::Tensor p, torch::Tensor g, double lr) {
+  check_f32_cuda(p, "param");
+  check_f32_cuda(g, "grad");
+  TORCH_CHECK(p.numel() == g.numel(), "size mismatch");
+  edl_dense_sgd(p.data_ptr<float>(), g.data_ptr<float>(), p.numel(),
+                static_cast<float>(lr), cur_stream());
+}
+
+void dense_momentum(torch::Tensor p, torch::Tensor vel, torch::Tensor g,
+                    double lr, double mu, bool nesterov) {
+  check_f32_cuda(p, "param");
+  edl_dense_momentum(p.data_ptr<float>(), vel.data_ptr<float>(),
+                     g.data_ptr<float>(), p.numel(), lr, mu, nesterov,
+                     cur_stream());
+}
+
+void dense_adam(torch::Tensor p, torch::Tensor m, torch::Tensor v,
+                c10::optional<torch::Tensor> max_sq, torch::Tensor g,
+                double lr_t, double b1, double b2, double eps) {
+  check_f32_cuda(p, "param");
+  edl_dense_adam(p.data_ptr<float>(), m.data_ptr<float>(), v.data_ptr<float>(),
+                 max_sq.has_value() ? max_sq->data_ptr<float>() : nullptr,
+                 g.data_ptr<float>(), p.numel(), lr_t, b1, b2, eps,
+                 cur_stream());
+}
+
+void dense_adagrad(torch::Tensor p, torch::Tensor m, torch::Tensor g,
+                   double lr, double eps) {
+  check_f32_cuda(p, "param");
+  edl_dense_adagrad(p.data_ptr<float>(), m.data_ptr<float>(),
+                    g.data_ptr<float>(), p.numel(), lr, eps, cur_stream());
+}
+
+void dense_ftrl(torch::Tensor p, torch::Tensor z, torch::Tensor n,
+                torch::Tensor g, double alpha, double beta, double l1,
+                double l2) {
+  check_f32_cuda(p, "param");
+  edl_dense_ftrl(p.data_ptr<float>(), z.data_ptr<float>(), n.data_ptr<float>(),
+                 g.data_ptr<float>(), p.numel(), alpha, beta, l1, l2,
+                 cur_stream());
+}
+
+// ------------------------------ sparse optimizers -----------------------
+static void check_sparse(const torch::Tensor& arena, const torch::Tensor& g,
+                         const torch::Tensor& slots) {
+  check_f32_cuda(arena, "arena");
+  check_f32_cuda(g, "grads");
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32, "slots must be int32");
+  TORCH_CHECK(g.dim() == 2 && arena.dim() == 2, "2-D tensors expected");
+  TORCH_CHECK(g.size(1) == arena.size(1), "dim mismatch");
+  TORCH_CHECK(slots.numel() == g.size(0), "slots/grads mismatch");
+}
+
+void sparse_sgd(torch::Tensor arena, torch::Tensor g, torch::Tensor slots,
+                double lr) {
+  check_sparse(arena, g, slots);
+  edl_sparse_sgd(arena.data_ptr<float>(), g.data_ptr<float>(),
+                 slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr,
+                 cur_stream());
+}
+
+void sparse_momentum(torch::Tensor arena, torch::Tensor vel, torch::Tensor g,
+                     torch::Tensor slots, double lr, double mu, bool nesterov) {
+  check_sparse(arena, g, slots);
+  edl_sparse_momentum(arena.data_ptr<float>(), vel.data_ptr<float>(),
+                      g.data_ptr<float>(), slots.data_ptr<int32_t>(),
+                      g.size(0), g.size(1), lr, mu, nesterov, cur_stream());
+}
+
+void sparse_adam(torch::Tensor arena, torch::Tensor m, torch::Tensor v,
+                 c10::optional<torch::Tensor> max_sq, torch::Tensor g,
+                 torch::Tensor slots, double lr_t, double b1, double b2,
+                 double eps) {
+  check_sparse(arena, g, slots);
+  edl_sparse_adam(arena.data_ptr<float>(), m.data_ptr<float>(),
+                  v.data_ptr<float>(),
+                  max_sq.has_value() ? max_sq->data_ptr<float>() : nullptr,
+                  g.data_ptr<float>(), slots.data_ptr<int32_t>(), g.size(0),
+                  g.size(1), lr_t, b1, b2, eps, cur_stream());
+}
+
+void sparse_adagrad(torch::Tensor arena, torch::Tensor m, torch::Tensor g,
+                    torch::Tensor slots, double lr, double eps) {
+  check_sparse(arena, g, slots);
+  edl_sparse_adagrad(arena.data_ptr<float>(), m.data_ptr<float>(),
+                     g.data_ptr<float>(), slots.data_ptr<int32_t>(), g.size(0),
+                     g.size(1), lr, eps, cur_stream());
+}
+
+void sparse_ftrl(torch::Tensor arena, torch::Tensor z, torch::Tensor n,
+                 torch::Tensor g, torch::Tensor slots, double alpha,
+                 double beta, double l1, double l2) {
+  check_sparse(arena, g, slots);
+  edl_sparse_ftrl(arena.data_ptr<float>(), z.data_ptr<float>(),
+                  n.data_ptr<float>(), g.data_ptr<float>(),
+                  slots.data_ptr<int32_t>(), g.size(0), g.size(1), alpha, beta,
+                  l1, l2, cur_stream());
+}
+
+// ------------------------------ hash table ------------------------------
+void ht_lookup_or_insert(torch::Tensor keys, torch::Tensor vals,
+                         torch::Tensor row_counter, int64_t max_rows,
+                         torch::Tensor ids, torch::Tensor out_slots,
+                         torch::Tensor out_is_new, torch::Tensor error_flag) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
+  TORCH_CHECK((keys.numel() & (keys.numel() - 1)) == 0,
+              "capacity must be a power of two");
+  TORCH_CHECK(ids.scalar_type() == torch::kInt64);
+  edl_ht_lookup_or_insert(
+      keys.data_ptr<int64_t>(), vals.data_ptr<int32_t>(), keys.numel(),
+      row_counter.data_ptr<int32_t>(), static_cast<int32_t>(max_rows),
+      ids.data_ptr<int64_t>(), ids.numel(), out_slots.data_ptr<int32_t>(),
+      out_is_new.data_ptr<uint8_t>(), error_flag.data_ptr<int32_t>(),
+      cur_stream());
+}
+
+void ht_lookup(torch::Tensor keys, torch::Tensor vals, torch::Tensor ids,
+               torch::Tensor out_slots) {
+  edl_ht_lookup(keys.data_ptr<int64_t>(), vals.data_ptr<int32_t>(),
+                keys.numel(), ids.data_ptr<int64_t>(), ids.numel(),
+                out_slots.data_ptr<int32_t>(), cur_stream());
+}
+
+void init_new_rows(torch::Tensor arena, torch::Tensor slots,
+                   torch::Tensor is_new, int64_t seed, double lo, double hi) {
+  check_f32_cuda(arena, "arena");
+  edl_init_new_rows(arena.data_ptr<float>(), slots.data_ptr<int32_t>(),
+                    is_new.data_ptr<uint8_t>(), slots.numel(), arena.size(1),
+                    static_cast<uint64_t>(seed), lo, hi, cur_stream());
+}
+
+torch::Tensor gather_rows(torch::Tensor arena, torch::Tensor slots) {
+  check_f32_cuda(arena, "arena");
+  auto out = torch::empty({slots.numel(), arena.size(1)}, arena.options());
+  edl_gather_rows(arena.data_ptr<float>(), slots.data_ptr<int32_t>(),
+                  slots.numel(), arena.size(1), out.data_ptr<float>(),
+                  cur_stream());
+  return out;
+}
+
+void scatter_rows(torch::Tensor arena, torch::Tensor slots,
+                  torch::Tensor rows) {
+  check_f32_cuda(arena, "arena");
+  check_f32_cuda(rows, "rows");
+  edl_scatter_rows(arena.data_ptr<float>(), slots.data_ptr<int32_t>(),
+                   rows.data_ptr<float>(), slots.numel(), arena.size(1),
+                   cur_stream());
+}
+
+// ------------------------------ fused GEMM ------------------------------
+// C[M,N] = act(A[M,K] @ B[N,K]^T + bias), bf16 in/out, f32 accumulate.
+torch::Tensor gemm_bias_act(torch::Tensor a, torch::Tensor b,
+                            c10::optional<torch::Tensor> bias, int64_t act) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), "GPU tensors expected");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  b.scalar_type() == torch::kBFloat16,
+              "bf16 expected");
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2);
+  TORCH_CHECK(a.size(1) == b.size(1), "K mismatch");
+  TORCH_CHECK(a.size(1) % 64 == 0, "K must be a multiple of 64 (pad)");
+  const float* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias must be f32");
+    TORCH_CHECK(bias->numel() == b.size(0));
+    bias_ptr = bias->data_ptr<float>();
+  }
+  auto c = torch::empty({a.size(0), b.size(0)}, a.options());
+  edl_gemm_bias_act_bf16(a.data_ptr(), b.data_ptr(), bias_ptr, c.data_ptr(),
+                         a.size(0), b.size(0), a.size(1),
+                         static_cast<int>(act), cur_stream());
+  return c;
+}
+
+// ---------------------- worker-side fused optimizers --------------------
+void fused_sgd_bf16(torch::Tensor p, torch::Tensor master, torch::Tensor vel,
+                    torch::Tensor g, double lr, double mu, bool nesterov,
+                    double weight_decay, double grad_scale) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(g.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(master.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(p.numel() == master.numel() && p.numel() == g.numel());
+  edl_fused_sgd_bf16(p.data_ptr(), master.data_ptr<float>(),
+                     vel.data_ptr<float>(), g.data_ptr(), p.numel(), lr, mu,
+                     nesterov, weight_decay, grad_scale, cur_stream());
+}
+
+void fused_adamw_bf16(torch::Tensor p, torch::Tensor master, torch::Tensor m,
+                      torch::Tensor v, torch::Tensor g, double lr_t, double b1,
+                      double b2, double eps, double weight_decay, double lr,
+                      double grad_scale) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(g.scalar_type() == torch::kBFloat16);
+  edl_fused_adamw_bf16(p.data_ptr(), master.data_ptr<float>(),
+                       m.data_ptr<float>(), v.data_ptr<float>(), g.data_ptr(),
+                       p.numel(), lr_t, b1, b2, eps, weight_decay, lr,
+                       grad_scale, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "elasticdl_amd MI355X kernels";
+  m.def("dense_sgd", &dense_sgd);
+  m.def("dense_momentum", &dense_momentum);
+  m.def("dense_adam", &dense_adam);
+  m.def("dense_adagrad", &dense_adagrad);
+  m.def("dense_ftrl", &dense_ftrl);
+  m.def("sparse_sgd", &sparse_sgd);
+  m.def("sparse_momentum", &sparse_momentum);
+  m.def("sparse_adam", &sparse_adam);
+  m.def("sparse_adagrad", &sparse_adagrad);
+  m.def("sparse_ftrl", &sparse_ftrl);
+  m.def("ht_lookup_or_insert", &ht_lookup_or_insert);
+  m.def("ht_lookup", &ht_lookup);
+  m.def("init_new_rows", &init_new_rows);
+  m.def("gather_rows", &gather_rows);
+  m.def("scatter_rows", &scatter_rows);
+  m.def("gemm_bias_act", &gemm_bias_act);
+  m.def("fused_sgd_bf16", &fused_sgd_bf16);
+  m.def("fused_adamw_bf16", &fused_adamw_bf16);
+}

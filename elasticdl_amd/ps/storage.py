@@ -1,0 +1,222 @@
+"""Embedding table storage: HBM-resident arena + GPU hash table.
+
+Replaces the reference's map-of-rows tables
+(elasticdl/go/pkg/common/embedding_table.go:21-88 — a Go map with a global
+RWMutex, one small allocation per row) with an MI355X-native layout:
+
+- one contiguous float32 arena [max_rows, dim] sized for HBM3E (288 GB/GPU
+  fits ~1e9 rows at dim 64 with Adam slots);
+- an open-addressing GPU hash table mapping id -> arena slot (HIP kernels
+  in ps_kernels.hip), so lookup/create of a whole batch of ids is one
+  kernel launch;
+- lazy row creation with deterministic on-device uniform init
+  (reference behavior: uniform(-0.05, 0.05) on first touch,
+  embedding_table.go:40-58);
+- slot arenas (optimizer state per row) allocated parallel to the main
+  arena and indexed by the same slots, mirroring the reference's
+  "<layer>-<slot>" tables (ps/parameters.py:169-183).
+
+On CPU (tests, local mode) the same API is served by a python dict +
+growable arena with bit-identical RNG init.
+"""
+
+from typing import Dict, Optional, Tuple
+
+import torch
+
+from elasticdl_amd.ops import reference, use_native
+
+
+def _next_pow2(n: int) -> int:
+    p = 1
+    while p < n:
+        p <<= 1
+    return p
+
+
+class EmbeddingTable:
+    def __init__(
+        self,
+        name: str,
+        dim: int,
+        device,
+        max_rows: int = 1 << 20,
+        initializer: Tuple[str, float, float] = ("uniform", -0.05, 0.05),
+        seed: int = 0x5EED,
+        is_slot: bool = False,
+        slot_init_value: float = 0.0,
+    ):
+        self.name = name
+        self.dim = dim
+        self.device = torch.device(device)
+        self.max_rows = max_rows
+        self.initializer = initializer
+        self.seed = seed
+        self.is_slot = is_slot
+        self.slot_init_value = slot_init_value
+        self._native = use_native(self.device)
+        self.slot_arenas: Dict[str, torch.Tensor] = {}
+
+        if self._native:
+            from elasticdl_amd.ops import _C
+
+            self._C = _C
+            cap = _next_pow2(max(2 * max_rows, 16))
+            self._keys = torch.full((cap,), -1, dtype=torch.int64, device=self.device)
+            self._vals = torch.zeros(cap, dtype=torch.int32, device=self.device)
+            self._counter = torch.zeros(1, dtype=torch.int32, device=self.device)
+            self._error = torch.zeros(1, dtype=torch.int32, device=self.device)
+            self._ids_by_slot = torch.full(
+                (max_rows,), -1, dtype=torch.int64, device=self.device
+            )
+            self.arena = torch.empty(
+                (max_rows, dim), dtype=torch.float32, device=self.device
+            )
+        else:
+            self._id_to_slot: Dict[int, int] = {}
+            self._grow = max(256, min(max_rows, 65536))
+            self.arena = torch.empty((self._grow, dim), dtype=torch.float32)
+            self._n_rows = 0
+
+    # ----------------------------------------------------------- properties
+    @property
+    def num_rows(self) -> int:
+        if self._native:
+            return int(self._counter.item())
+        return self._n_rows
+
+    def check_health(self) -> None:
+        if self._native and int(self._error.item()) != 0:
+            raise RuntimeError(
+                f"EmbeddingTable {self.name}: arena/hash-table overflow "
+                f"(max_rows={self.max_rows})"
+            )
+
+    # -------------------------------------------------------------- slots
+    def get_slot_arena(self, slot_name: str) -> torch.Tensor:
+        """Optimizer state arena parallel to the main arena."""
+        arena = self.slot_arenas.get(slot_name)
+        if arena is None:
+            arena = torch.zeros_like(self.arena)
+            self.slot_arenas[slot_name] = arena
+        elif arena.shape[0] < self.arena.shape[0]:  # CPU arena grew
+            extra = torch.zeros(
+                (self.arena.shape[0] - arena.shape[0], self.dim),
+                dtype=arena.dtype,
+            )
+            arena = torch.cat([arena, extra], dim=0)
+            self.slot_arenas[slot_name] = arena
+        return arena
+
+    # ------------------------------------------------------------- lookups
+    def lookup_or_create(self, unique_ids: torch.Tensor) -> torch.Tensor:
+        """Map unique ids -> arena slots, lazily creating + initializing new
+        rows. Returns int32 slots on the table's device."""
+        unique_ids = unique_ids.to(self.device, torch.int64)
+        if self._native:
+            n = unique_ids.numel()
+            slots = torch.empty(n, dtype=torch.int32, device=self.device)
+            is_new = torch.empty(n, dtype=torch.uint8, device=self.device)
+            self._C.ht_lookup_or_insert(
+                self._keys,
+                self._vals,
+                self._counter,
+                self.max_rows,
+                unique_ids,
+                slots,
+                is_new,
+                self._error,
+            )
+            lo, hi = self._init_range()
+            self._C.init_new_rows(self.arena, slots, is_new, self.seed, lo, hi)
+            # remember id per slot for checkpoint export
+            new_mask = is_new.bool()
+            if bool(new_mask.any()):
+                self._ids_by_slot.index_copy_(
+                    0, slots[new_mask].long(), unique_ids[new_mask]
+                )
+            return slots
+        # ----- CPU path
+        slots = torch.empty(unique_ids.numel(), dtype=torch.int32)
+        new_slots = []
+        for i, v in enumerate(unique_ids.tolist()):
+            s = self._id_to_slot.get(v)
+            if s is None:
+                s = self._n_rows
+                if s >= self.max_rows:
+                    raise RuntimeError(f"EmbeddingTable {self.name} full")
+                self._id_to_slot[v] = s
+                self._n_rows += 1
+                self._ensure_capacity(self._n_rows)
+                new_slots.append(s)
+            slots[i] = s
+        if new_slots:
+            ns = torch.tensor(new_slots, dtype=torch.int32)
+            lo, hi = self._init_range()
+            self.arena[ns.long()] = reference.init_rows_values(
+                ns, self.dim, self.seed, lo, hi
+            )
+        return slots
+
+    def _init_range(self) -> Tuple[float, float]:
+        if self.is_slot:
+            return (self.slot_init_value, self.slot_init_value)
+        kind, lo, hi = self.initializer
+        return (lo, hi)
+
+    def lookup(self, ids: torch.Tensor) -> torch.Tensor:
+        """Read-only lookup: slot or -1 per id."""
+        ids = ids.to(self.device, torch.int64)
+        if self._native:
+            out = torch.empty(ids.numel(), dtype=torch.int32, device=self.device)
+            self._C.ht_lookup(self._keys, self._vals, ids, out)
+            return out
+        return torch.tensor(
+            [self._id_to_slot.get(v, -1) for v in ids.tolist()],
+            dtype=torch.int32,
+        )
+
+    def gather(self, ids: torch.Tensor, create: bool = True) -> torch.Tensor:
+        """Rows for (possibly duplicate) ids; missing rows are created
+        (training) or zero (create=False)."""
+        ids = ids.to(self.device, torch.int64)
+        unique_ids, inverse = torch.unique(ids, sorted=True, return_inverse=True)
+        if create:
+            slots = self.lookup_or_create(unique_ids)
+        else:
+            slots = self.lookup(unique_ids)
+        full_slots = slots.index_select(0, inverse.view(-1).to(slots.device))
+        if self._native:
+            return self._C.gather_rows(self.arena, full_slots)
+        return reference.gather_rows(self.arena, full_slots)
+
+    # ----------------------------------------------------------- checkpoint
+    def export_rows(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        """(ids [n], rows [n, dim]) of all live rows, on CPU."""
+        if self._native:
+            n = self.num_rows
+            ids = self._ids_by_slot[:n].cpu()
+            rows = self.arena[:n].cpu()
+            return ids.clone(), rows.clone()
+        n = self._n_rows
+        ids = torch.empty(n, dtype=torch.int64)
+        for v, s in self._id_to_slot.items():
+            ids[s] = v
+        return ids, self.arena[:n].clone()
+
+    def import_rows(self, ids: torch.Tensor, rows: torch.Tensor) -> None:
+        slots = self.lookup_or_create(ids.to(torch.int64))
+        rows = rows.to(self.device, torch.float32).contiguous()
+        if self._native:
+            self._C.scatter_rows(self.arena, slots, rows)
+        else:
+            reference.scatter_rows(self.arena, slots, rows)
+
+    # ----------------------------------------------------------------- util
+    def _ensure_capacity(self, rows_needed: int) -> None:
+        if self.arena.shape[0] < rows_needed:
+            grow = max(self._grow, rows_needed - self.arena.shape[0])
+            self.arena = torch.cat(
+                [self.arena, torch.empty((grow, self.dim), dtype=torch.float32)],
+                dim=0,
+            )

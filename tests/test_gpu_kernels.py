@@ -1,0 +1,256 @@
+"""HIP kernel numerics vs the pure-torch fp32 reference (GPU only).
+
+Mirrors the reference's golden kernel tests (kernel_test.go) with the CPU
+implementations in elasticdl_amd.ops.reference as the oracle.
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from elasticdl_amd.ops import reference  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def C():
+    from elasticdl_amd.ops import require_native
+
+    return require_native()
+
+
+def rand(*shape):
+    return torch.randn(*shape, dtype=torch.float32)
+
+
+# --------------------------- dense optimizers ----------------------------
+@pytest.mark.parametrize("numel", [16, 1023, 100003])
+def test_dense_sgd(C, numel):
+    p = rand(numel)
+    g = rand(numel)
+    p_gpu = p.cuda()
+    C.dense_sgd(p_gpu, g.cuda(), 0.1)
+    reference.dense_sgd(p, g, 0.1)
+    torch.cuda.synchronize()
+    assert torch.allclose(p_gpu.cpu(), p, atol=1e-6)
+
+
+def test_dense_momentum(C):
+    p, v, g = rand(5000), rand(5000), rand(5000)
+    pg, vg = p.cuda(), v.cuda()
+    for _ in range(3):
+        C.dense_momentum(pg, vg, g.cuda(), 0.1, 0.9, True)
+        reference.dense_momentum(p, v, g, 0.1, 0.9, True)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-5)
+    assert torch.allclose(vg.cpu(), v, atol=1e-5)
+
+
+@pytest.mark.parametrize("amsgrad", [False, True])
+def test_dense_adam(C, amsgrad):
+    n = 4097
+    p, m, v, g = rand(n), torch.zeros(n), torch.zeros(n), rand(n)
+    ms = torch.zeros(n) if amsgrad else None
+    pg, mg, vg = p.cuda(), m.cuda(), v.cuda()
+    msg = ms.cuda() if amsgrad else None
+    for step in range(1, 4):
+        lr_t = reference.adam_lr_t(0.01, step, 0.9, 0.999)
+        C.dense_adam(pg, mg, vg, msg, g.cuda(), lr_t, 0.9, 0.999, 1e-8)
+        reference.dense_adam(p, m, v, ms, g, lr_t, 0.9, 0.999, 1e-8)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-5)
+
+
+def test_dense_adagrad(C):
+    p, m, g = rand(1000), torch.zeros(1000), rand(1000)
+    pg, mg = p.cuda(), m.cuda()
+    C.dense_adagrad(pg, mg, g.cuda(), 0.1, 1e-7)
+    reference.dense_adagrad(p, m, g, 0.1, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-6)
+
+
+def test_dense_ftrl(C):
+    n = 1000
+    p, z, acc, g = rand(n), rand(n), rand(n).abs(), rand(n)
+    pg, zg, ag = p.cuda(), z.cuda(), acc.cuda()
+    C.dense_ftrl(pg, zg, ag, g.cuda(), 0.5, 1.0, 0.01, 0.01)
+    reference.dense_ftrl(p, z, acc, g, 0.5, 1.0, 0.01, 0.01)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-5)
+    assert torch.allclose(zg.cpu(), z, atol=1e-5)
+
+
+# --------------------------- sparse optimizers ---------------------------
+@pytest.mark.parametrize("dim", [4, 8, 64, 63])
+def test_sparse_sgd(C, dim):
+    arena = rand(100, dim)
+    g = rand(10, dim)
+    slots = torch.randperm(100)[:10].to(torch.int32)
+    ag = arena.cuda()
+    C.sparse_sgd(ag, g.cuda(), slots.cuda(), 0.1)
+    reference.sparse_sgd(arena, g, slots, 0.1)
+    torch.cuda.synchronize()
+    assert torch.allclose(ag.cpu(), arena, atol=1e-6)
+
+
+def test_sparse_adam(C):
+    dim, n = 16, 20
+    arena, m, v = rand(100, dim), torch.zeros(100, dim), torch.zeros(100, dim)
+    g = rand(n, dim)
+    slots = torch.randperm(100)[:n].to(torch.int32)
+    ag, mg, vg = arena.cuda(), m.cuda(), v.cuda()
+    lr_t = reference.adam_lr_t(0.01, 1, 0.9, 0.999)
+    C.sparse_adam(ag, mg, vg, None, g.cuda(), slots.cuda(), lr_t, 0.9, 0.999, 1e-8)
+    reference.sparse_adam(arena, m, v, None, g, slots, lr_t, 0.9, 0.999, 1e-8)
+    torch.cuda.synchronize()
+    assert torch.allclose(ag.cpu(), arena, atol=1e-6)
+    assert torch.allclose(mg.cpu(), m, atol=1e-6)
+
+
+def test_sparse_ftrl(C):
+    dim, n = 8, 15
+    arena, z, acc = rand(50, dim), rand(50, dim), rand(50, dim).abs()
+    g = rand(n, dim)
+    slots = torch.randperm(50)[:n].to(torch.int32)
+    ag, zg, ng = arena.cuda(), z.cuda(), acc.cuda()
+    C.sparse_ftrl(ag, zg, ng, g.cuda(), slots.cuda(), 0.5, 1.0, 0.01, 0.01)
+    reference.sparse_ftrl(arena, z, acc, g, slots, 0.5, 1.0, 0.01, 0.01)
+    torch.cuda.synchronize()
+    assert torch.allclose(ag.cpu(), arena, atol=1e-5)
+
+
+# ------------------------------ hash table -------------------------------
+def _make_ht(C, cap=1 << 16, max_rows=1 << 15):
+    keys = torch.full((cap,), -1, dtype=torch.int64, device="cuda")
+    vals = torch.zeros(cap, dtype=torch.int32, device="cuda")
+    counter = torch.zeros(1, dtype=torch.int32, device="cuda")
+    err = torch.zeros(1, dtype=torch.int32, device="cuda")
+    return keys, vals, counter, err, max_rows
+
+
+def test_ht_insert_lookup_bulk(C):
+    keys, vals, counter, err, max_rows = _make_ht(C)
+    ids = torch.randperm(1 << 20)[: 10000].to(torch.int64).cuda()
+    slots = torch.empty(10000, dtype=torch.int32, device="cuda")
+    is_new = torch.empty(10000, dtype=torch.uint8, device="cuda")
+    C.ht_lookup_or_insert(keys, vals, counter, max_rows, ids, slots, is_new, err)
+    torch.cuda.synchronize()
+    assert int(err.item()) == 0
+    assert int(counter.item()) == 10000
+    assert bool(is_new.all())
+    # all slots distinct, in [0, 10000)
+    s = slots.cpu()
+    assert s.min() >= 0 and s.max() < 10000
+    assert s.unique().numel() == 10000
+    # second call: same slots, nothing new
+    slots2 = torch.empty_like(slots)
+    C.ht_lookup_or_insert(keys, vals, counter, max_rows, ids, slots2, is_new, err)
+    torch.cuda.synchronize()
+    assert torch.equal(slots, slots2)
+    assert not bool(is_new.any())
+    assert int(counter.item()) == 10000
+    # read-only lookup: found + missing
+    probe = torch.cat([ids[:5].cpu(), torch.tensor([(1 << 21) + 1, (1 << 21) + 2])])
+    out = torch.empty(7, dtype=torch.int32, device="cuda")
+    C.ht_lookup(keys, vals, probe.cuda(), out)
+    torch.cuda.synchronize()
+    assert torch.equal(out[:5].cpu(), slots[:5].cpu())
+    assert out[5] == -1 and out[6] == -1
+
+
+def test_ht_arena_full_sets_error(C):
+    keys, vals, counter, err, _ = _make_ht(C, cap=256, max_rows=10)
+    ids = torch.arange(20, dtype=torch.int64).cuda()
+    slots = torch.empty(20, dtype=torch.int32, device="cuda")
+    is_new = torch.empty(20, dtype=torch.uint8, device="cuda")
+    C.ht_lookup_or_insert(keys, vals, counter, 10, ids, slots, is_new, err)
+    torch.cuda.synchronize()
+    assert int(err.item()) == 1
+
+
+# -------------------------- gather / init / scatter ----------------------
+def test_gather_rows(C):
+    arena = rand(64, 16).cuda()
+    slots = torch.tensor([3, -1, 7, 3], dtype=torch.int32).cuda()
+    out = C.gather_rows(arena, slots)
+    torch.cuda.synchronize()
+    ref = reference.gather_rows(arena.cpu(), slots.cpu())
+    assert torch.equal(out.cpu(), ref)
+
+
+def test_init_rows_bit_identical_to_cpu(C):
+    dim, seed = 8, 12345
+    arena = torch.zeros(32, dim, device="cuda")
+    slots = torch.arange(10, dtype=torch.int32, device="cuda")
+    is_new = torch.ones(10, dtype=torch.uint8, device="cuda")
+    C.init_new_rows(arena, slots, is_new, seed, -0.05, 0.05)
+    torch.cuda.synchronize()
+    ref = reference.init_rows_values(slots.cpu(), dim, seed, -0.05, 0.05)
+    assert torch.allclose(arena[:10].cpu(), ref, atol=0), (
+        arena[:10].cpu() - ref
+    ).abs().max()
+
+
+def test_scatter_rows(C):
+    arena = torch.zeros(32, 8, device="cuda")
+    slots = torch.tensor([5, 9], dtype=torch.int32).cuda()
+    rows = rand(2, 8).cuda()
+    C.scatter_rows(arena, slots, rows)
+    torch.cuda.synchronize()
+    assert torch.equal(arena[5].cpu(), rows[0].cpu())
+    assert torch.equal(arena[9].cpu(), rows[1].cpu())
+
+
+# ------------------------------- fused GEMM ------------------------------
+@pytest.mark.parametrize(
+    "M,N,K", [(128, 128, 64), (256, 512, 128), (100, 200, 64), (513, 300, 192)]
+)
+def test_gemm_matches_matmul(C, M, N, K):
+    torch.manual_seed(0)
+    a = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+    out = C.gemm_bias_act(a, b, None, 0)
+    torch.cuda.synchronize()
+    ref = (a.float() @ b.float().t())
+    # bf16 inputs, f32 accumulate, bf16 store
+    assert torch.allclose(out.float(), ref, atol=2e-1, rtol=2e-2), (
+        (out.float() - ref).abs().max()
+    )
+
+
+def test_gemm_bias_relu(C):
+    torch.manual_seed(1)
+    a = torch.randn(64, 64, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(32, 64, dtype=torch.bfloat16, device="cuda")
+    bias = torch.randn(32, dtype=torch.float32, device="cuda")
+    out = C.gemm_bias_act(a, b, bias, 1)
+    torch.cuda.synchronize()
+    ref = torch.relu(a.float() @ b.float().t() + bias)
+    assert torch.allclose(out.float(), ref, atol=2e-1, rtol=2e-2)
+
+
+def test_engine_gpu_matches_cpu_end_to_end():
+    """Whole PSEngine step on GPU vs CPU (same seed -> identical init)."""
+    from elasticdl_amd.common.tensor_utils import IndexedSlices
+    from elasticdl_amd.ps.engine import PSEngine
+
+    def run(device):
+        e = PSEngine(
+            opt_type="adam",
+            opt_args="learning_rate=0.01",
+            device=device,
+            use_async=True,
+            seed=99,
+        )
+        e.push_model({"w": torch.ones(8)}, [{"name": "emb", "dim": 8}])
+        ids = torch.tensor([3, 5, 3], dtype=torch.int64)
+        rows = e.pull_embedding_vectors("emb", ids)
+        g = IndexedSlices(torch.ones(3, 8), ids)
+        e.push_gradients({"w": torch.ones(8)}, {"emb": g}, version=0)
+        return rows.cpu(), e.pull_embedding_vectors("emb", ids).cpu(), e.dense["w"].cpu()
+
+    r_cpu = run("cpu")
+    r_gpu = run("cuda")
+    for c, g in zip(r_cpu, r_gpu):
+        assert torch.allclose(c, g, atol=1e-6), (c - g).abs().max()
