@@ -60,6 +60,7 @@ def bench_resnet50(args, rank, world, local_rank):
     device = torch.device("cuda", local_rank) if torch.cuda.is_available() else torch.device("cpu")
     if device.type == "cuda":
         torch.cuda.set_device(device)
+        torch.backends.cudnn.benchmark = True  # let MIOpen autotune convs
 
     torch.manual_seed(1234)
     model = resnet.resnet50(num_classes=args.num_classes)

@@ -52,6 +52,25 @@ def feed(batch, device, dtype=None):
     return images.to(device), labels.to(device)
 
 
+def custom_data_reader(data_origin: str = ""):
+    """Synthetic MNIST-shaped reader (no network -> no real dataset);
+    data_origin may be 'synthetic:<size>'."""
+    from elasticdl_amd.data.reader import SyntheticReader
+
+    size = 640
+    if data_origin.startswith("synthetic:"):
+        size = int(data_origin.split(":", 1)[1])
+
+    def sample(i: int):
+        g = torch.Generator().manual_seed(i)
+        return (
+            torch.randn(1, 28, 28, generator=g),
+            torch.randint(0, 10, (1,), generator=g)[0],
+        )
+
+    return SyntheticReader(size, sample, name="mnist-synthetic")
+
+
 def synthetic_batch(batch_size: int = 64, seed: int = None):
     g = torch.Generator().manual_seed(seed) if seed is not None else None
     return (

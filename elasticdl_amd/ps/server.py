@@ -1,0 +1,229 @@
+"""Parameter-server daemon: gRPC surface over the PSEngine.
+
+The rebuild of the Go PS daemon (elasticdl/go/cmd/elasticdl_ps/main.go +
+go/pkg/ps/server.go:53-253): same RPC surface (push_model,
+push_embedding_table_infos, pull_dense_parameters, pull_embedding_vectors,
+push_gradients) on the raw-bytes codec, backed by the GPU engine. Extras:
+- version -> master report every evaluation_steps (server.go:122-126);
+- checkpoint every checkpoint_steps versions (server.go:128-141);
+- exits when the master is finished (polled; main.go:59-72).
+"""
+
+import argparse
+import os
+import threading
+import time
+from typing import Optional
+
+import torch
+
+from elasticdl_amd.common import rpc
+from elasticdl_amd.common.log_utils import default_logger as logger
+from elasticdl_amd.common.tensor_utils import IndexedSlices
+from elasticdl_amd.ps.engine import PSEngine
+
+
+class PserverServicer:
+    def __init__(
+        self,
+        engine: PSEngine,
+        master_client=None,
+        evaluation_steps: int = 0,
+        checkpoint_dir: Optional[str] = None,
+        checkpoint_steps: int = 0,
+        keep_checkpoint_max: int = 3,
+    ):
+        self.engine = engine
+        self._master_client = master_client
+        self._evaluation_steps = evaluation_steps
+        self._checkpoint_dir = checkpoint_dir
+        self._checkpoint_steps = checkpoint_steps
+        self._keep_checkpoint_max = keep_checkpoint_max
+        engine.version_listeners.append(self._on_version)
+
+    def methods(self):
+        return {
+            "push_model": self.push_model,
+            "push_embedding_table_infos": self.push_embedding_table_infos,
+            "pull_dense_parameters": self.pull_dense_parameters,
+            "pull_embedding_vectors": self.pull_embedding_vectors,
+            "push_gradients": self.push_gradients,
+        }
+
+    # ------------------------------------------------------------ handlers
+    def push_model(self, req: dict) -> dict:
+        accepted = self.engine.push_model(
+            req.get("dense_parameters", {}),
+            req.get("embedding_table_infos", []),
+        )
+        return {"accepted": accepted, "version": self.engine.version}
+
+    def push_embedding_table_infos(self, req: dict) -> dict:
+        self.engine.push_embedding_table_infos(req.get("infos", []))
+        return {}
+
+    def pull_dense_parameters(self, req: dict) -> dict:
+        ok, version, params = self.engine.pull_dense(req.get("version", -1))
+        resp = {"initialized": ok, "version": version}
+        if params is not None:
+            resp["dense_parameters"] = params
+        return resp
+
+    def pull_embedding_vectors(self, req: dict) -> dict:
+        ids = req["ids"].to(torch.int64)
+        rows = self.engine.pull_embedding_vectors(
+            req["name"], ids, create=req.get("create", True)
+        )
+        return {"rows": rows.cpu()}
+
+    def push_gradients(self, req: dict) -> dict:
+        embedding = {
+            name: IndexedSlices(d["values"], d["ids"].to(torch.int64))
+            for name, d in req.get("embedding_gradients", {}).items()
+        }
+        accepted, version = self.engine.push_gradients(
+            req.get("dense_gradients", {}),
+            embedding,
+            learning_rate=req.get("learning_rate", 1.0),
+            version=req.get("version", 0),
+        )
+        return {"accepted": accepted, "version": version}
+
+    # ------------------------------------------------------------- hooks
+    def _on_version(self, version: int) -> None:
+        if (
+            self._evaluation_steps
+            and version % self._evaluation_steps == 0
+            and self._master_client is not None
+        ):
+            try:
+                self._master_client.report_version(version)
+            except Exception:  # noqa: BLE001 - master may be restarting
+                logger.warning("report_version(%d) failed", version)
+        if (
+            self._checkpoint_steps
+            and self._checkpoint_dir
+            and version % self._checkpoint_steps == 0
+        ):
+            from elasticdl_amd.utils.save_utils import CheckpointSaver
+
+            CheckpointSaver(
+                self._checkpoint_dir, keep_max=self._keep_checkpoint_max
+            ).save_shard(
+                version,
+                self.engine.state_for_checkpoint(),
+                self.engine.shard_id,
+                self.engine.num_shards,
+            )
+
+
+class ParameterServer:
+    """Process wrapper: engine + gRPC server + master liveness polling."""
+
+    def __init__(self, args: argparse.Namespace):
+        self.args = args
+        device = args.device
+        if device == "auto":
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.engine = PSEngine(
+            shard_id=args.ps_id,
+            num_shards=args.num_ps_pods,
+            opt_type=args.opt_type,
+            opt_args=args.opt_args,
+            device=device,
+            use_async=args.use_async,
+            grads_to_wait=args.grads_to_wait,
+            lr_staleness_modulation=args.lr_staleness_modulation,
+            sync_version_tolerance=args.sync_version_tolerance,
+            embedding_max_rows=args.embedding_max_rows,
+        )
+        if args.checkpoint_dir_for_init:
+            from elasticdl_amd.utils.save_utils import CheckpointSaver
+
+            state = CheckpointSaver.load_for_shard(
+                args.checkpoint_dir_for_init, args.ps_id, args.num_ps_pods
+            )
+            if state is not None:
+                self.engine.restore_from_checkpoint(state)
+                logger.info(
+                    "PS %d restored from %s (version %d)",
+                    args.ps_id,
+                    args.checkpoint_dir_for_init,
+                    self.engine.version,
+                )
+        self._master_client = None
+        if args.master_addr:
+            from elasticdl_amd.worker.master_client import MasterClient
+
+            self._master_client = MasterClient(args.master_addr, worker_id=-1)
+        self.servicer = PserverServicer(
+            self.engine,
+            master_client=self._master_client,
+            evaluation_steps=args.evaluation_steps,
+            checkpoint_dir=args.checkpoint_dir,
+            checkpoint_steps=args.checkpoint_steps,
+            keep_checkpoint_max=args.keep_checkpoint_max,
+        )
+        self.server = None
+        self._stop = threading.Event()
+
+    def start(self) -> int:
+        self.server = rpc.start_server(
+            f"0.0.0.0:{self.args.port}",
+            {"Pserver": self.servicer.methods()},
+            max_workers=min(max(self.args.num_workers, 1), 64),
+        )
+        logger.info("PS %d serving on port %d", self.args.ps_id, self.server.port)
+        return self.server.port
+
+    def run(self) -> None:
+        """Serve until the master reports the job finished."""
+        self.start()
+        try:
+            while not self._stop.wait(30.0):
+                if self._master_client is not None:
+                    try:
+                        if self._master_client.job_finished():
+                            logger.info("Master finished; PS exiting")
+                            break
+                    except Exception:  # noqa: BLE001
+                        logger.warning("master liveness check failed")
+        finally:
+            self.server.stop(2)
+
+    def stop(self) -> None:
+        self._stop.set()
+
+
+def parse_ps_args(argv=None) -> argparse.Namespace:
+    """Flag mirror of the Go PS CLI (go/cmd/elasticdl_ps/main.go:27-74)."""
+    p = argparse.ArgumentParser("elasticdl_ps")
+    p.add_argument("--port", type=int, default=0)
+    p.add_argument("--ps_id", type=int, default=0)
+    p.add_argument("--num_ps_pods", type=int, default=1)
+    p.add_argument("--num_workers", type=int, default=2)
+    p.add_argument("--opt_type", default="sgd")
+    p.add_argument("--opt_args", default="learning_rate=0.1")
+    p.add_argument("--use_async", type=lambda s: s.lower() == "true", default=True)
+    p.add_argument("--grads_to_wait", type=int, default=1)
+    p.add_argument(
+        "--lr_staleness_modulation", type=lambda s: s.lower() == "true", default=False
+    )
+    p.add_argument("--sync_version_tolerance", type=int, default=0)
+    p.add_argument("--evaluation_steps", type=int, default=0)
+    p.add_argument("--checkpoint_dir", default="")
+    p.add_argument("--checkpoint_steps", type=int, default=0)
+    p.add_argument("--keep_checkpoint_max", type=int, default=3)
+    p.add_argument("--checkpoint_dir_for_init", default="")
+    p.add_argument("--embedding_max_rows", type=int, default=1 << 22)
+    p.add_argument("--device", default="auto")
+    p.add_argument("--master_addr", default=os.environ.get("EDL_MASTER_ADDR", ""))
+    return p.parse_args(argv)
+
+
+def main(argv=None):
+    ParameterServer(parse_ps_args(argv)).run()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,103 @@
+"""Model-zoo loading.
+
+Mirrors common/model_utils.py:27-242: a model definition is a Python
+module (file path or dotted module name) exposing the zoo contract —
+``custom_model()``, ``loss()``, ``optimizer()``, ``eval_metrics_fn()``,
+``feed()`` and optionally ``custom_data_reader()`` / ``callbacks()`` /
+``synthetic_batch()``. Built-in zoo modules resolve by short name
+(mnist, resnet50, wide_deep, deepfm, ...).
+"""
+
+import importlib
+import importlib.util
+import os
+from dataclasses import dataclass, field
+from typing import Callable, Optional
+
+_BUILTIN = {
+    "mnist": "elasticdl_amd.models.mnist",
+    "resnet50": "elasticdl_amd.models.resnet",
+    "wide_deep": "elasticdl_amd.models.wide_deep",
+    "deepfm": "elasticdl_amd.models.deepfm",
+}
+
+
+def load_module(model_def: str):
+    if model_def in _BUILTIN:
+        return importlib.import_module(_BUILTIN[model_def])
+    if os.path.isfile(model_def):
+        spec = importlib.util.spec_from_file_location(
+            os.path.splitext(os.path.basename(model_def))[0], model_def
+        )
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        return mod
+    return importlib.import_module(model_def)
+
+
+@dataclass
+class ModelSpec:
+    module: object
+    model_fn: Callable
+    loss_fn: Callable
+    optimizer_fn: Callable
+    feed_fn: Optional[Callable] = None
+    eval_metrics_fn: Optional[Callable] = None
+    data_reader_fn: Optional[Callable] = None
+    callbacks_fn: Optional[Callable] = None
+    params: dict = field(default_factory=dict)
+
+    def build_model(self):
+        return self.model_fn(**self.params)
+
+
+def get_model_spec(model_def: str, model_params: Optional[dict] = None) -> ModelSpec:
+    mod = load_module(model_def)
+
+    def need(name):
+        fn = getattr(mod, name, None)
+        if fn is None:
+            raise AttributeError(
+                f"model zoo module {model_def!r} must define {name}()"
+            )
+        return fn
+
+    return ModelSpec(
+        module=mod,
+        model_fn=need("custom_model"),
+        loss_fn=need("loss"),
+        optimizer_fn=need("optimizer"),
+        feed_fn=getattr(mod, "feed", None),
+        eval_metrics_fn=getattr(mod, "eval_metrics_fn", None),
+        data_reader_fn=getattr(mod, "custom_data_reader", None),
+        callbacks_fn=getattr(mod, "callbacks", None),
+        params=model_params or {},
+    )
+
+
+def get_optimizer_info(opt) -> tuple:
+    """Normalize a zoo optimizer() return into (opt_type, opt_args) for the
+    PS CLI (reference: get_optimizer_info, common/model_utils.py:227)."""
+    if isinstance(opt, tuple):
+        return opt
+    import torch
+
+    if isinstance(opt, torch.optim.SGD):
+        g = opt.param_groups[0]
+        return (
+            "momentum" if g.get("momentum", 0) else "sgd",
+            f"learning_rate={g['lr']};momentum={g.get('momentum', 0)}"
+            f";nesterov={str(bool(g.get('nesterov'))).lower()}",
+        )
+    if isinstance(opt, torch.optim.Adam):
+        g = opt.param_groups[0]
+        b1, b2 = g["betas"]
+        return (
+            "adam",
+            f"learning_rate={g['lr']};beta_1={b1};beta_2={b2}"
+            f";epsilon={g['eps']}",
+        )
+    if isinstance(opt, torch.optim.Adagrad):
+        g = opt.param_groups[0]
+        return ("adagrad", f"learning_rate={g['lr']};epsilon={g['eps']}")
+    raise ValueError(f"cannot map optimizer {type(opt)} to PS opt_type")

@@ -1,0 +1,180 @@
+"""Elastic AllReduce trainer (RCCL over xGMI).
+
+Rebuild of elasticdl/python/worker/allreduce_trainer.py:37-146 +
+elasticai_api/pytorch/controller.py:97-203 on the framework's own stack:
+
+- CommunicatorManager polls the master; when the rendezvous generation
+  changes, the RCCL process group is torn down and re-formed, and rank 0
+  re-broadcasts model + optimizer state + the completed-batch counter;
+- collective failures (a peer died mid-allreduce) sleep 3 s, re-init, and
+  the minibatch retries (<=5 attempts, reference :66-91);
+- fixed global batch: backward_passes_per_step is recomputed from the
+  current world size every step so adding/removing workers keeps the
+  global batch constant (controller.py:178-203).
+"""
+
+import time
+from typing import Optional
+
+import torch
+
+from elasticdl_amd.collective.communicator import CommunicatorManager
+from elasticdl_amd.collective.distributed_optimizer import DistributedOptimizer
+from elasticdl_amd.common.constants import MAX_ALLREDUCE_RETRY_NUM
+from elasticdl_amd.common.log_utils import default_logger as logger
+from elasticdl_amd.utils.model_utils import ModelSpec
+from elasticdl_amd.worker.trainer import Trainer
+
+_WORLD_CHECK_INTERVAL_SEC = 20.0
+
+
+class AllReduceTrainer(Trainer):
+    def __init__(
+        self,
+        spec: ModelSpec,
+        master_client,
+        device: str = "cpu",
+        lr: float = 0.1,
+        momentum: float = 0.9,
+        global_batch_num_per_step: Optional[int] = None,
+        bucket_cap_mb: float = 25.0,
+    ):
+        self.spec = spec
+        self.mc = master_client
+        self.device = torch.device(device)
+        dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        self.dtype = dtype
+        model = spec.build_model().to(self.device, dtype)
+        if self.device.type == "cuda":
+            model = model.to(memory_format=torch.channels_last)
+            torch.backends.cudnn.benchmark = True
+        self.model = model
+        self.comm = CommunicatorManager(
+            master_client=master_client, worker_host=master_client.worker_host
+        )
+        self.opt = DistributedOptimizer(
+            model, lr=lr, momentum=momentum, bucket_cap_mb=bucket_cap_mb
+        )
+        self.global_batch_num_per_step = global_batch_num_per_step
+        self._completed_batches = 0
+        self._last_world_check = 0.0
+        self._version = 0
+
+    # ---------------------------------------------------------- elasticity
+    def init_communicator_if_needed(self, force: bool = False) -> None:
+        now = time.monotonic()
+        if not force and now - self._last_world_check < _WORLD_CHECK_INTERVAL_SEC \
+                and self.comm.rendezvous_id >= 0 and self.comm.world_size > 0:
+            return
+        self._last_world_check = now
+        reformed = self.comm.ensure_communicator()
+        if reformed or self.comm.need_broadcast:
+            self._broadcast_state()
+            self.comm.need_broadcast = False
+        self._adjust_accumulation()
+
+    def _broadcast_state(self) -> None:
+        if self.comm.world_size <= 1:
+            return
+        import torch.distributed as dist
+
+        logger.info(
+            "Broadcasting model from rank 0 (world=%d, gen=%d)",
+            self.comm.world_size,
+            self.comm.rendezvous_id,
+        )
+        for b in self.opt.buckets:
+            if b.param_flat is not None:
+                dist.broadcast(b.param_flat, 0)
+                b.master.copy_(b.param_flat.float())
+            else:
+                for p in b.params:
+                    dist.broadcast(p.data, 0)
+        for t in self.model.buffers():
+            if t.numel() > 0 and t.dtype.is_floating_point:
+                dist.broadcast(t.data, 0)
+        self._completed_batches = int(
+            self.comm.broadcast_value(float(self._completed_batches), 0)
+        )
+
+    def _adjust_accumulation(self) -> None:
+        """Fixed global batch across world resizes (controller.py:178-203)."""
+        if not self.global_batch_num_per_step:
+            return
+        world = max(1, self.comm.world_size)
+        rank = max(0, self.comm.rank)
+        n = self.global_batch_num_per_step // world
+        if rank < self.global_batch_num_per_step % world:
+            n += 1
+        self.opt.set_backward_passes_per_step(max(1, n))
+
+    # ------------------------------------------------------------ training
+    def train_minibatch(self, batch):
+        for attempt in range(MAX_ALLREDUCE_RETRY_NUM + 1):
+            try:
+                self.init_communicator_if_needed(force=attempt > 0)
+                return self._train_once(batch)
+            except RuntimeError as e:
+                msg = str(e).lower()
+                if any(k in msg for k in ("nccl", "rccl", "connection",
+                                          "timeout", "process group")):
+                    logger.warning(
+                        "Collective failed (%s); re-initializing (%d/%d)",
+                        e, attempt + 1, MAX_ALLREDUCE_RETRY_NUM,
+                    )
+                    self.comm.teardown()
+                    time.sleep(3)
+                    continue
+                raise
+        raise RuntimeError("allreduce retries exhausted")
+
+    def _train_once(self, batch):
+        x, y = self._feed(batch)
+        self.opt.zero_grad()
+        out = self.model(x)
+        loss = self.spec.loss_fn(out.float(), y)
+        loss.backward()
+        step_due = self.opt.record_backward_pass()
+        if step_due:
+            self.opt.step()
+            self._version += 1
+        self._completed_batches += 1
+        return loss.detach(), self._version
+
+    def _feed(self, batch):
+        if self.spec.feed_fn is not None:
+            try:
+                return self.spec.feed_fn(batch, self.device, self.dtype)
+            except TypeError:
+                return self.spec.feed_fn(batch, self.device)
+        x, y = batch
+        return x.to(self.device, self.dtype), y.to(self.device)
+
+    @torch.no_grad()
+    def evaluate_minibatch(self, batch):
+        x, y = self._feed(batch)
+        return self.model(x).float(), y
+
+    @torch.no_grad()
+    def predict_minibatch(self, batch):
+        x, _ = self._feed(batch)
+        return self.model(x).float()
+
+    def get_model_version(self) -> int:
+        return self._version
+
+    def export_model(self, path: str) -> None:
+        if self.comm.rank <= 0:
+            torch.save(self.model.state_dict(), path)
+            logger.info("Exported model to %s", path)
+
+    def on_training_end(self) -> None:
+        from elasticdl_amd.master.servicer import TrainingLoopStatus
+
+        self.mc.report_training_loop_status(TrainingLoopStatus.END)
+        self.comm.teardown()
+
+    def on_training_start(self) -> None:
+        from elasticdl_amd.master.servicer import TrainingLoopStatus
+
+        self.mc.report_training_loop_status(TrainingLoopStatus.START)

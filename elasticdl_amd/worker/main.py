@@ -1,0 +1,84 @@
+"""Worker entrypoint (reference: worker/main.py:40-62): reads
+EDL_WORKER_ID / EDL_MASTER_ADDR from env (overridable by flags), builds
+the master client + trainer for the distribution strategy, runs the
+task loop."""
+
+import os
+import sys
+
+import torch
+
+from elasticdl_amd.common.args import parse_model_params, parse_worker_args
+from elasticdl_amd.common.constants import DistributionStrategy, WorkerEnv
+from elasticdl_amd.common.log_utils import default_logger as logger
+from elasticdl_amd.utils.model_utils import get_model_spec
+from elasticdl_amd.worker.master_client import MasterClient
+from elasticdl_amd.worker.worker import Worker
+
+
+def build_worker(args) -> Worker:
+    worker_id = args.worker_id
+    if worker_id < 0:
+        worker_id = int(os.environ.get(WorkerEnv.WORKER_ID, 0))
+    master_addr = args.master_addr or os.environ.get(WorkerEnv.MASTER_ADDR, "")
+    ps_addrs = args.ps_addrs or os.environ.get(WorkerEnv.PS_ADDRS, "")
+
+    device = args.device
+    if device == "auto":
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    spec = get_model_spec(args.model_def, parse_model_params(args.model_params))
+    mc = MasterClient(master_addr, worker_id)
+
+    reader = None
+    if spec.data_reader_fn is not None:
+        reader = spec.data_reader_fn(args.training_data)
+    elif args.training_data:
+        from elasticdl_amd.data.reader import create_data_reader
+
+        reader = create_data_reader(args.training_data)
+
+    if args.distribution_strategy == DistributionStrategy.PARAMETER_SERVER:
+        from elasticdl_amd.worker.ps_client import PSClient
+        from elasticdl_amd.worker.ps_trainer import ParameterServerTrainer
+
+        assert ps_addrs, "PS strategy needs --ps_addrs"
+        trainer = ParameterServerTrainer(
+            spec,
+            PSClient(ps_addrs.split(",")),
+            device=device,
+            get_model_steps=args.get_model_steps,
+            use_async=args.use_async,
+        )
+    elif args.distribution_strategy == DistributionStrategy.ALLREDUCE:
+        from elasticdl_amd.worker.allreduce_trainer import AllReduceTrainer
+
+        trainer = AllReduceTrainer(spec, mc, device=device)
+    else:
+        from elasticdl_amd.worker.trainer import LocalTrainer
+
+        trainer = LocalTrainer(spec, device=device)
+
+    return Worker(
+        worker_id=worker_id,
+        master_client=mc,
+        trainer=trainer,
+        data_reader=reader,
+        spec=spec,
+        minibatch_size=args.minibatch_size,
+        log_loss_steps=args.log_loss_steps,
+        export_path=args.output,
+    )
+
+
+def main(argv=None) -> int:
+    args = parse_worker_args(argv)
+    worker = build_worker(args)
+    logger.info("Worker %d starting (strategy=%s)",
+                worker.worker_id, args.distribution_strategy)
+    worker.run()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

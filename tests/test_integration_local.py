@@ -1,0 +1,197 @@
+"""In-process distributed integration tests (the reference's primary
+harness pattern: real gRPC master + PS + worker in one process,
+test_utils.py:330-460), plus the full subprocess local mode."""
+
+import os
+import subprocess
+import sys
+import tempfile
+
+import pytest
+import torch
+
+from elasticdl_amd.common import rpc
+from elasticdl_amd.common.task import TaskType
+from elasticdl_amd.master.evaluation_service import EvaluationService
+from elasticdl_amd.master.servicer import MasterServicer
+from elasticdl_amd.master.task_manager import TaskManager
+from elasticdl_amd.ps.server import ParameterServer, parse_ps_args
+from elasticdl_amd.utils.model_utils import get_model_spec
+from elasticdl_amd.worker.master_client import MasterClient
+from elasticdl_amd.worker.ps_client import PSClient
+from elasticdl_amd.worker.ps_trainer import ParameterServerTrainer
+from elasticdl_amd.worker.worker import Worker
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def start_master(spec, reader, records_per_task=32, **tm_kw):
+    tm = TaskManager(
+        training_shards=reader.create_shards(),
+        records_per_task=records_per_task,
+        **tm_kw,
+    )
+    ev = EvaluationService(tm, metrics_fn=spec.eval_metrics_fn)
+    servicer = MasterServicer(tm, evaluation_service=ev)
+    server = rpc.start_server("127.0.0.1:0", {"Master": servicer.methods()})
+    return tm, ev, servicer, server
+
+
+def start_ps(num_ps, opt_type="sgd", opt_args="learning_rate=0.1", **kw):
+    servers = []
+    addrs = []
+    for i in range(num_ps):
+        argv = [
+            "--port", "0", "--ps_id", str(i), "--num_ps_pods", str(num_ps),
+            "--opt_type", opt_type, "--opt_args", opt_args, "--device", "cpu",
+        ]
+        for k, v in kw.items():
+            argv += [f"--{k}", str(v)]
+        ps = ParameterServer(parse_ps_args(argv))
+        port = ps.start()
+        servers.append(ps)
+        addrs.append(f"127.0.0.1:{port}")
+    return servers, addrs
+
+
+def test_mnist_ps_training_in_process():
+    spec = get_model_spec("mnist")
+    reader = spec.data_reader_fn("synthetic:128")
+    tm, ev, servicer, server = start_master(spec, reader)
+    ps_servers, ps_addrs = start_ps(2)
+    try:
+        mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=0)
+        trainer = ParameterServerTrainer(spec, PSClient(ps_addrs), device="cpu")
+        worker = Worker(0, mc, trainer, data_reader=reader, spec=spec,
+                        minibatch_size=16)
+        worker.run()
+        assert tm.finished()
+        assert tm.completed_steps == 4  # 128 / 32
+        assert trainer.get_model_version() > 0
+        # dense params actually live on the PS shards (2 shards)
+        total = sum(len(ps.engine.dense) for ps in ps_servers)
+        n_params = len([p for p in trainer.model.parameters() if p.requires_grad])
+        assert total == n_params
+    finally:
+        server.stop(0)
+        for ps in ps_servers:
+            ps.server.stop(0)
+
+
+def test_wide_deep_ps_training_with_embeddings():
+    spec = get_model_spec("wide_deep",)
+    from elasticdl_amd.data.reader import SyntheticReader
+    from elasticdl_amd.models import wide_deep
+
+    def sample(i):
+        ids, labels = wide_deep.synthetic_batch(1, num_features=13, vocab=1000,
+                                                seed=i)
+        return ids[0], labels[0]
+
+    reader = SyntheticReader(96, sample)
+    tm, ev, servicer, server = start_master(spec, reader, records_per_task=48)
+    ps_servers, ps_addrs = start_ps(2, opt_type="adam",
+                                    opt_args="learning_rate=0.001")
+    try:
+        mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=0)
+        trainer = ParameterServerTrainer(spec, PSClient(ps_addrs), device="cpu")
+        worker = Worker(0, mc, trainer, data_reader=reader, spec=spec,
+                        minibatch_size=16)
+        worker.run()
+        assert tm.finished()
+        # embedding rows created on both shards
+        rows = [
+            ps.engine.tables["deep_embedding"].num_rows for ps in ps_servers
+        ]
+        assert all(r > 0 for r in rows), rows
+    finally:
+        server.stop(0)
+        for ps in ps_servers:
+            ps.server.stop(0)
+
+
+def test_two_workers_share_tasks():
+    import threading
+
+    spec = get_model_spec("mnist")
+    reader = spec.data_reader_fn("synthetic:256")
+    tm, ev, servicer, server = start_master(spec, reader)
+    ps_servers, ps_addrs = start_ps(1)
+    try:
+        def run_worker(wid):
+            mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=wid)
+            trainer = ParameterServerTrainer(spec, PSClient(ps_addrs),
+                                             device="cpu")
+            Worker(wid, mc, trainer, data_reader=reader, spec=spec,
+                   minibatch_size=16).run()
+
+        threads = [threading.Thread(target=run_worker, args=(i,)) for i in range(2)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(180)
+        assert tm.finished()
+        assert tm.completed_steps == 8
+    finally:
+        server.stop(0)
+        for ps in ps_servers:
+            ps.server.stop(0)
+
+
+def test_sync_sgd_two_workers_grads_to_wait():
+    spec = get_model_spec("mnist")
+    reader = spec.data_reader_fn("synthetic:64")
+    tm, ev, servicer, server = start_master(spec, reader)
+    ps_servers, ps_addrs = start_ps(1, use_async="false", grads_to_wait=2,
+                                    sync_version_tolerance=2)
+    try:
+        import threading
+
+        def run_worker(wid):
+            mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=wid)
+            trainer = ParameterServerTrainer(spec, PSClient(ps_addrs),
+                                             device="cpu")
+            Worker(wid, mc, trainer, data_reader=reader, spec=spec,
+                   minibatch_size=16).run()
+
+        threads = [threading.Thread(target=run_worker, args=(i,)) for i in range(2)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(180)
+        assert tm.finished()
+        # sync PS: version == pushes / grads_to_wait (4 tasks x 2... but
+        # workers race; just require it advanced and is consistent)
+        assert ps_servers[0].engine.version >= 1
+    finally:
+        server.stop(0)
+        for ps in ps_servers:
+            ps.server.stop(0)
+
+
+@pytest.mark.timeout(300)
+def test_local_mode_subprocess_e2e():
+    """BASELINE config 1: master + 1 worker + 1 PS as real processes."""
+    with tempfile.TemporaryDirectory() as tmp:
+        export = os.path.join(tmp, "model.pt")
+        cmd = [
+            sys.executable, "-m", "elasticdl_amd.master.main",
+            "--model_def", "mnist",
+            "--distribution_strategy", "ParameterServerStrategy",
+            "--num_workers", "1",
+            "--num_ps_pods", "1",
+            "--minibatch_size", "16",
+            "--num_minibatches_per_task", "2",
+            "--training_data", "synthetic:96",
+            "--device", "cpu",
+            "--checkpoint_dir", tmp,
+            "--output", export,
+            "--pod_manager", "local",
+        ]
+        env = dict(os.environ, PYTHONPATH=REPO)
+        r = subprocess.run(cmd, env=env, cwd=REPO, capture_output=True,
+                           text=True, timeout=280)
+        assert r.returncode == 0, r.stderr[-3000:]
+        assert os.path.exists(export), "train-end export missing"
+        state = torch.load(export, weights_only=True)
+        assert any("net" in k for k in state)
