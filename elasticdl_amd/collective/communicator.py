@@ -30,14 +30,20 @@ class CommunicatorManager:
         master_client=None,
         worker_host: str = "",
         backend: Optional[str] = None,
-        init_timeout: float = 300.0,
+        init_timeout: float = None,
     ):
         self._master_client = master_client
         self._worker_host = worker_host
         self._backend = backend or (
             "nccl" if torch.cuda.is_available() else "gloo"
         )
+        if init_timeout is None:
+            init_timeout = float(os.environ.get("EDL_PG_TIMEOUT_SEC", "60"))
         self._init_timeout = init_timeout
+        # a peer dying mid-collective must surface as an exception in the
+        # survivors (re-init path), not kill their processes:
+        # 2 = CleanUpOnly (abort the RCCL communicator, raise from work)
+        os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "2")
         self.rendezvous_id = -1
         self.rank = -1
         self.world_size = 0

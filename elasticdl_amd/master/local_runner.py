@@ -81,6 +81,9 @@ class LocalProcessManager:
         proc = LocalProcess(name, popen, pod_type, pod_id)
         with self._lock:
             self.procs[name] = proc
+        if self.log_dir:
+            with open(os.path.join(self.log_dir, f"{name}.pid"), "w") as f:
+                f.write(str(popen.pid))
         logger.info("Started %s (pid %d)", name, popen.pid)
         return proc
 

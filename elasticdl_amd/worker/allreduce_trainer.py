@@ -25,7 +25,7 @@ from elasticdl_amd.common.log_utils import default_logger as logger
 from elasticdl_amd.utils.model_utils import ModelSpec
 from elasticdl_amd.worker.trainer import Trainer
 
-_WORLD_CHECK_INTERVAL_SEC = 20.0
+_WORLD_CHECK_INTERVAL_SEC = 5.0
 
 
 class AllReduceTrainer(Trainer):
