@@ -1,0 +1,280 @@
+"""Kubernetes client wrapper: pod CRUD + label patch + event watch.
+
+Rebuild of the reference's two k8s clients
+(elasticdl_client/common/k8s_client.py:220-410 — pod/service CRUD,
+labels, owner references; elasticdl/python/common/k8s_client.py:41-111 —
+the master-side event-watch thread). The ``kubernetes`` package import is
+deferred so the rest of the framework works without it; tests inject a
+fake core API.
+
+Labels (reference: k8s_client.py:29-33):
+    elasticdl-job-name, elasticdl-replica-type, elasticdl-replica-index
+"""
+
+import threading
+import traceback
+from typing import Callable, Dict, List, Optional
+
+from elasticdl_amd.common.log_utils import default_logger as logger
+
+ELASTICDL_JOB_KEY = "elasticdl-job-name"
+ELASTICDL_REPLICA_TYPE_KEY = "elasticdl-replica-type"
+ELASTICDL_REPLICA_INDEX_KEY = "elasticdl-replica-index"
+
+
+def parse_resource(spec: str) -> Dict[str, str]:
+    """'cpu=4,memory=8192Mi,amd.com/gpu=1' -> k8s resource dict
+    (reference: elasticdl_client/common/k8s_resource.py)."""
+    out: Dict[str, str] = {}
+    for kv in (spec or "").split(","):
+        kv = kv.strip()
+        if not kv:
+            continue
+        k, _, v = kv.partition("=")
+        k = k.strip()
+        if k == "gpu":
+            k = "amd.com/gpu"
+        out[k] = v.strip()
+    return out
+
+
+def parse_volume(spec: str) -> List[Dict[str, str]]:
+    """'claim_name=x,mount_path=/data;...' -> volume dicts
+    (reference: k8s_volume.py)."""
+    out = []
+    for part in (spec or "").split(";"):
+        part = part.strip()
+        if not part:
+            continue
+        vol = {}
+        for kv in part.split(","):
+            k, _, v = kv.partition("=")
+            vol[k.strip()] = v.strip()
+        out.append(vol)
+    return out
+
+
+class Client:
+    def __init__(
+        self,
+        namespace: str,
+        job_name: str,
+        image_name: str = "",
+        core_api=None,
+        force_use_kube_config_file: bool = False,
+    ):
+        self.namespace = namespace
+        self.job_name = job_name
+        self.image_name = image_name
+        if core_api is not None:
+            self.client = core_api
+        else:
+            self.client = self._create_core_api(force_use_kube_config_file)
+
+    @staticmethod
+    def _create_core_api(use_kube_config: bool):
+        from kubernetes import client, config
+
+        try:
+            if use_kube_config:
+                config.load_kube_config()
+            else:
+                config.load_incluster_config()
+        except Exception:  # noqa: BLE001 - fall back to kubeconfig
+            config.load_kube_config()
+        return client.CoreV1Api()
+
+    # ------------------------------------------------------------- naming
+    def get_master_pod_name(self) -> str:
+        return f"elasticdl-{self.job_name}-master"
+
+    def get_pod_name(self, pod_type: str, index: int) -> str:
+        return f"elasticdl-{self.job_name}-{pod_type}-{index}"
+
+    # --------------------------------------------------------------- CRUD
+    def create_pod(self, pod) -> bool:
+        try:
+            self.client.create_namespaced_pod(self.namespace, pod)
+            return True
+        except Exception:  # noqa: BLE001
+            logger.warning("create_pod failed:\n%s", traceback.format_exc())
+            return False
+
+    def delete_pod(self, pod_name: str) -> bool:
+        try:
+            from kubernetes import client as k8s
+
+            self.client.delete_namespaced_pod(
+                pod_name,
+                self.namespace,
+                body=k8s.V1DeleteOptions(grace_period_seconds=0),
+            )
+            return True
+        except ImportError:
+            self.client.delete_namespaced_pod(pod_name, self.namespace)
+            return True
+        except Exception:  # noqa: BLE001
+            logger.warning("delete_pod(%s) failed", pod_name)
+            return False
+
+    def get_pod(self, pod_name: str):
+        try:
+            return self.client.read_namespaced_pod(pod_name, self.namespace)
+        except Exception:  # noqa: BLE001
+            return None
+
+    def patch_labels_to_pod(self, pod_name: str, labels: Dict[str, str]):
+        try:
+            return self.client.patch_namespaced_pod(
+                pod_name, self.namespace, {"metadata": {"labels": labels}}
+            )
+        except Exception:  # noqa: BLE001
+            logger.warning("patch labels on %s failed", pod_name)
+            return None
+
+    # -------------------------------------------------------------- specs
+    def build_pod_spec(
+        self,
+        pod_name: str,
+        pod_type: str,
+        index: int,
+        command: List[str],
+        resource_requests: str,
+        resource_limits: str = "",
+        priority_class: str = "",
+        envs: Optional[Dict[str, str]] = None,
+        volumes: str = "",
+        image_pull_policy: str = "IfNotPresent",
+        restart_policy: str = "Never",
+        owner_pod=None,
+    ):
+        """V1Pod with the elasticdl labels + owner reference to the master
+        pod (reference: k8s_client.py:283-298)."""
+        from kubernetes import client as k8s
+
+        env = [k8s.V1EnvVar(name=k, value=str(v)) for k, v in (envs or {}).items()]
+        env.append(
+            k8s.V1EnvVar(
+                name="MY_POD_IP",
+                value_from=k8s.V1EnvVarSource(
+                    field_ref=k8s.V1ObjectFieldSelector(field_path="status.podIP")
+                ),
+            )
+        )
+        requests = parse_resource(resource_requests)
+        limits = parse_resource(resource_limits) or requests
+        volume_mounts = []
+        pod_volumes = []
+        for i, vol in enumerate(parse_volume(volumes)):
+            name = vol.get("name", f"edl-volume-{i}")
+            if "claim_name" in vol:
+                pod_volumes.append(
+                    k8s.V1Volume(
+                        name=name,
+                        persistent_volume_claim=(
+                            k8s.V1PersistentVolumeClaimVolumeSource(
+                                claim_name=vol["claim_name"]
+                            )
+                        ),
+                    )
+                )
+            elif "host_path" in vol:
+                pod_volumes.append(
+                    k8s.V1Volume(
+                        name=name,
+                        host_path=k8s.V1HostPathVolumeSource(
+                            path=vol["host_path"]
+                        ),
+                    )
+                )
+            volume_mounts.append(
+                k8s.V1VolumeMount(name=name, mount_path=vol["mount_path"])
+            )
+        container = k8s.V1Container(
+            name="main",
+            image=self.image_name,
+            command=command,
+            resources=k8s.V1ResourceRequirements(
+                requests=requests, limits=limits
+            ),
+            env=env,
+            volume_mounts=volume_mounts or None,
+            image_pull_policy=image_pull_policy,
+        )
+        spec = k8s.V1PodSpec(
+            containers=[container],
+            restart_policy=restart_policy,
+            priority_class_name=priority_class or None,
+            volumes=pod_volumes or None,
+        )
+        owner_refs = None
+        if owner_pod is not None:
+            owner_refs = [
+                k8s.V1OwnerReference(
+                    api_version="v1",
+                    kind="Pod",
+                    name=owner_pod.metadata.name,
+                    uid=owner_pod.metadata.uid,
+                    block_owner_deletion=True,
+                    controller=True,
+                )
+            ]
+        return k8s.V1Pod(
+            api_version="v1",
+            kind="Pod",
+            metadata=k8s.V1ObjectMeta(
+                name=pod_name,
+                labels={
+                    "app": "elasticdl",
+                    ELASTICDL_JOB_KEY: self.job_name,
+                    ELASTICDL_REPLICA_TYPE_KEY: pod_type,
+                    ELASTICDL_REPLICA_INDEX_KEY: str(index),
+                },
+                owner_references=owner_refs,
+            ),
+            spec=spec,
+        )
+
+    # -------------------------------------------------------------- watch
+    def start_watch(self, event_callback: Callable, periodic_callback:
+                    Optional[Callable] = None, interval: float = 15.0):
+        """Label-selected pod event watch in a daemon thread
+        (reference: common/k8s_client.py:92-106) plus an optional periodic
+        callback (retry pod creation etc.)."""
+
+        def watch_loop():
+            from kubernetes import watch
+
+            while True:
+                try:
+                    stream = watch.Watch().stream(
+                        self.client.list_namespaced_pod,
+                        self.namespace,
+                        label_selector=f"{ELASTICDL_JOB_KEY}={self.job_name}",
+                    )
+                    for event in stream:
+                        event_callback(event)
+                except Exception:  # noqa: BLE001 - watch reconnects
+                    logger.warning("k8s watch reconnecting:\n%s",
+                                   traceback.format_exc())
+                import time
+
+                time.sleep(2)
+
+        t = threading.Thread(target=watch_loop, name="k8s-watch", daemon=True)
+        t.start()
+        if periodic_callback is not None:
+            def periodic_loop():
+                import time
+
+                while True:
+                    time.sleep(interval)
+                    try:
+                        periodic_callback()
+                    except Exception:  # noqa: BLE001
+                        logger.warning("periodic callback failed")
+
+            threading.Thread(
+                target=periodic_loop, name="k8s-periodic", daemon=True
+            ).start()
+        return t
