@@ -1,0 +1,11 @@
+from elasticdl_amd.preprocessing.layers import (  # noqa: F401
+    ConcatenateWithOffset,
+    Discretization,
+    Hashing,
+    IndexLookup,
+    LogRound,
+    Normalizer,
+    RoundIdentity,
+    SparseEmbedding,
+    ToNumber,
+)
