@@ -4,9 +4,13 @@ equivalents on MI355X. Prints one JSON line per benchmark."""
 
 import argparse
 import json
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def timeit(fn, warmup=10, iters=50):
