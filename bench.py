@@ -242,7 +242,7 @@ def main():
     args = ap.parse_args()
 
     if args.batch_size is None:
-        args.batch_size = 256 if args.model == "resnet50" else 4096
+        args.batch_size = 512 if args.model == "resnet50" else 4096
         if not torch.cuda.is_available():
             args.batch_size = 16
 

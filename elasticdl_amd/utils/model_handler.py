@@ -1,0 +1,87 @@
+"""Strategy-specific model rewrite.
+
+Mirrors the reference ModelHandler (common/model_handler.py:148-461): for
+the ParameterServer strategy, every ``nn.Embedding`` whose weight exceeds
+a size threshold (reference: 2 MB, model_handler.py:78-103) is replaced
+in-place by a PS-backed EdlEmbedding of the same dim; the inverse rewrite
+materializes trained PS rows back into stock ``nn.Embedding`` layers for
+export.
+"""
+
+from typing import Dict, List, Tuple
+
+import torch
+import torch.nn as nn
+
+from elasticdl_amd.common.constants import DistributionStrategy
+from elasticdl_amd.layers.embedding import EdlEmbedding
+
+_DEFAULT_THRESHOLD_BYTES = 2 * 1024 * 1024
+
+
+class ModelHandler:
+    @staticmethod
+    def get_model_handler(distribution_strategy: str, **kw):
+        if distribution_strategy == DistributionStrategy.PARAMETER_SERVER:
+            return ParameterServerModelHandler(**kw)
+        return DefaultModelHandler()
+
+
+class DefaultModelHandler(ModelHandler):
+    def get_model_to_train(self, model: nn.Module) -> nn.Module:
+        return model
+
+    def get_model_to_export(self, model: nn.Module, engine=None) -> nn.Module:
+        return model
+
+
+class ParameterServerModelHandler(ModelHandler):
+    def __init__(self, threshold_bytes: int = _DEFAULT_THRESHOLD_BYTES):
+        self.threshold_bytes = threshold_bytes
+        self._replaced: List[Tuple[str, int, int]] = []  # (name, rows, dim)
+
+    def get_model_to_train(self, model: nn.Module) -> nn.Module:
+        """Replace big nn.Embedding layers by EdlEmbedding in place."""
+        for parent, attr, qualified, child in _named_children_deep(model):
+            if isinstance(child, nn.Embedding):
+                size = child.weight.numel() * child.weight.element_size()
+                if size >= self.threshold_bytes:
+                    edl = EdlEmbedding(
+                        name=qualified,
+                        dim=child.embedding_dim,
+                        max_rows=max(child.num_embeddings * 2, 1024),
+                    )
+                    setattr(parent, attr, edl)
+                    self._replaced.append(
+                        (qualified, child.num_embeddings, child.embedding_dim)
+                    )
+        return model
+
+    def get_model_to_export(self, model: nn.Module, engine=None) -> nn.Module:
+        """Inverse rewrite: EdlEmbedding -> nn.Embedding with rows pulled
+        from the PS engine/client (reference: model_handler.py:242-285)."""
+        for parent, attr, qualified, child in _named_children_deep(model):
+            if isinstance(child, EdlEmbedding) and engine is not None:
+                rows_spec = next(
+                    (r for r in self._replaced if r[0] == child.name), None
+                )
+                num_rows = rows_spec[1] if rows_spec else child.max_rows
+                emb = nn.Embedding(num_rows, child.dim)
+                ids = torch.arange(num_rows, dtype=torch.int64)
+                with torch.no_grad():
+                    rows = engine.pull_embedding_vectors(
+                        child.name, ids, create=False
+                    )
+                    emb.weight.copy_(rows.cpu())
+                setattr(parent, attr, emb)
+        return model
+
+
+def _named_children_deep(model: nn.Module):
+    """(parent, attr, qualified_name, child) for every submodule."""
+    out = []
+    for parent_name, parent in model.named_modules():
+        for attr, child in parent.named_children():
+            qualified = f"{parent_name}.{attr}" if parent_name else attr
+            out.append((parent, attr, qualified, child))
+    return out
