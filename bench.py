@@ -72,8 +72,12 @@ def bench_resnet50(args, rank, world, local_rank):
     )
     if world > 1:
         for b in opt.buckets:  # one broadcast per flat buffer, rank-0 state
-            dist.broadcast(b.param_flat, 0)
-            b.master.copy_(b.param_flat.float())
+            if b.param_flat is not None:
+                dist.broadcast(b.param_flat, 0)
+                b.master.copy_(b.param_flat.float())
+            else:  # CPU fallback path keeps params in place
+                for p in b.params:
+                    dist.broadcast(p.data, 0)
 
     bs = args.batch_size
     n_batches = 4  # rotate a few resident synthetic batches
@@ -134,18 +138,22 @@ def bench_resnet50(args, rank, world, local_rank):
 
 
 def bench_ps_model(args, rank, world, local_rank):
-    """PS path: DeepFM/Wide&Deep with a colocated GPU PS engine per rank
-    (embedding shard in local HBM; dense towers via the MFMA GEMM)."""
+    """PS path: DeepFM/Wide&Deep with the embedding table sharded across
+    the ranks' HBM (id % world), rows exchanged over RCCL all-to-all on
+    xGMI; dense towers run data-parallel through the bucketed
+    DistributedOptimizer + MFMA fused GEMMs."""
+    from elasticdl_amd.collective.distributed_optimizer import DistributedOptimizer
     from elasticdl_amd.common.tensor_utils import merge_indexed_slices
-    from elasticdl_amd.layers.embedding import bind_local_engine, find_edl_embeddings
+    from elasticdl_amd.layers.embedding import find_edl_embeddings
     from elasticdl_amd.models import deepfm, wide_deep
     from elasticdl_amd.ps.engine import PSEngine
+    from elasticdl_amd.ps.sharded import ShardedPSEngine
 
     dist = setup_dist(world)
     device = torch.device("cuda", local_rank) if torch.cuda.is_available() else torch.device("cpu")
     if device.type == "cuda":
         torch.cuda.set_device(device)
-    torch.manual_seed(1234 + rank)
+    torch.manual_seed(1234)  # same dense init on all ranks
 
     zoo = deepfm if args.model == "deepfm" else wide_deep
     if args.model == "deepfm":
@@ -155,33 +163,35 @@ def bench_ps_model(args, rank, world, local_rank):
         model = wide_deep.WideDeep(max_rows=args.table_rows)
         batch_fn = lambda s: wide_deep.synthetic_batch(args.batch_size, seed=s)
     model = model.to(device)
+    if device.type == "cuda":
+        model = model.to(torch.bfloat16)
 
     opt_type, opt_args = zoo.optimizer()
-    engine = PSEngine(
+    local_engine = PSEngine(
+        shard_id=rank, num_shards=world,
         opt_type=opt_type, opt_args=opt_args,
         device=device, use_async=True,
-        embedding_max_rows=args.table_rows,
+        embedding_max_rows=max(args.table_rows // max(world, 1), 1024),
     )
-    engine.push_model(
-        {},
-        [e.table_info() for e in find_edl_embeddings(model)],
-    )
-    bind_local_engine(model, engine)
+    embeddings = find_edl_embeddings(model)
+    local_engine.push_model({}, [e.table_info() for e in embeddings])
+    engine = ShardedPSEngine(local_engine)
     sink = []
-    for e in find_edl_embeddings(model):
+    for e in embeddings:
+        e.lookup_fn = lambda name, ids: engine.pull_embedding_vectors(name, ids)
         e.set_grad_sink(sink)
 
-    dense_opt = torch.optim.Adam(
-        [p for p in model.parameters() if p.requires_grad], lr=1e-3
+    dense_opt = DistributedOptimizer(
+        model, opt_type="adamw", lr=1e-3, bucket_cap_mb=args.bucket_mb
     )
 
-    batches = [batch_fn(s) for s in range(4)]
+    batches = [batch_fn(1000 * rank + s) for s in range(4)]
     batches = [(ids.to(device), y.to(device)) for ids, y in batches]
 
     def one_step(i):
         ids, y = batches[i % len(batches)]
         sink.clear()
-        dense_opt.zero_grad(set_to_none=True)
+        dense_opt.zero_grad()
         out = model(ids)
         loss = zoo.loss(out, y)
         loss.backward()
@@ -190,7 +200,7 @@ def bench_ps_model(args, rank, world, local_rank):
         for n, s in sink:
             by_name.setdefault(n, []).append(s)
         merged = {n: merge_indexed_slices(*lst) for n, lst in by_name.items()}
-        engine.push_gradients({}, merged, version=0)
+        engine.push_sparse_gradients(merged, version=0)
         return loss
 
     for i in range(args.warmup):
