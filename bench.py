@@ -40,7 +40,8 @@ def sync_all(dist, world, device):
         torch.cuda.synchronize(device)
     if world > 1:
         dist.barrier()
-        torch.cuda.synchronize(device)
+        if torch.cuda.is_available():
+            torch.cuda.synchronize(device)
 
 
 def max_over_ranks(dist, world, value: float, device) -> float:
