@@ -31,7 +31,7 @@ _MASTER_ARG_KEYS = {
     "image_name", "worker_resource_request", "worker_resource_limit",
     "ps_resource_request", "ps_resource_limit", "worker_pod_priority",
     "ps_pod_priority", "volume", "image_pull_policy", "restart_policy",
-    "relaunch_on_worker_failure", "num_minibatches_per_shard",
+    "relaunch_on_worker_failure", "num_minibatches_per_shard", "job_type",
 }
 
 
@@ -43,8 +43,8 @@ def build_master_command(args) -> List[str]:
 
 
 def submit_job(args, job_type: str) -> int:
-    if job_type == "evaluate":
-        args.num_epochs = 0
+    if job_type in ("evaluate", "predict"):
+        args.job_type = job_type
     if not args.image_name:
         # no image -> run the master locally (single-node mode)
         logger.info("No --image_name: running master locally")

@@ -66,10 +66,26 @@ class Master:
                 args.validation_data
             ).create_shards()
 
+        prediction_shards = None
+        if args.prediction_data:
+            from elasticdl_amd.data.reader import create_data_reader
+
+            prediction_shards = create_data_reader(
+                args.prediction_data
+            ).create_shards()
+
+        # job type derivation (reference: elasticdl_job_service.py:32-54)
+        self.job_type = args.job_type or (
+            "predict" if prediction_shards and not training_shards
+            else "evaluate" if evaluation_shards and not training_shards
+            else "train"
+        )
+
         records_per_task = args.minibatch_size * args.num_minibatches_per_task
         self.task_manager = TaskManager(
-            training_shards=training_shards,
+            training_shards=training_shards if self.job_type == "train" else None,
             evaluation_shards=evaluation_shards,
+            prediction_shards=prediction_shards,
             records_per_task=records_per_task,
             num_epochs=args.num_epochs,
             max_step=args.max_step,
@@ -77,6 +93,10 @@ class Master:
             shuffle_shards=args.shuffle_shards,
             task_timeout_sec=args.task_timeout_sec,
         )
+        if self.job_type == "evaluate":
+            self.task_manager.create_evaluation_tasks(model_version=0)
+        elif self.job_type == "predict":
+            self.task_manager.create_prediction_tasks()
         if args.output:
             self.task_manager.enable_train_end_callback()
         if args.checkpoint_dir_for_init:
@@ -129,6 +149,7 @@ class Master:
             "--minibatch_size", str(a.minibatch_size),
             "--get_model_steps", str(a.get_model_steps),
             "--training_data", a.training_data,
+            "--validation_data", a.validation_data,
             "--device", a.device,
             "--log_loss_steps", str(a.log_loss_steps),
         ]

@@ -31,12 +31,18 @@ def build_worker(args) -> Worker:
     mc = MasterClient(master_addr, worker_id)
 
     reader = None
+    eval_reader = None
     if spec.data_reader_fn is not None:
         reader = spec.data_reader_fn(args.training_data)
-    elif args.training_data:
+        if args.validation_data:
+            eval_reader = spec.data_reader_fn(args.validation_data)
+    else:
         from elasticdl_amd.data.reader import create_data_reader
 
-        reader = create_data_reader(args.training_data)
+        if args.training_data:
+            reader = create_data_reader(args.training_data)
+        if args.validation_data:
+            eval_reader = create_data_reader(args.validation_data)
 
     if args.distribution_strategy == DistributionStrategy.PARAMETER_SERVER:
         from elasticdl_amd.worker.ps_client import PSClient
@@ -64,6 +70,7 @@ def build_worker(args) -> Worker:
         master_client=mc,
         trainer=trainer,
         data_reader=reader,
+        eval_data_reader=eval_reader,
         spec=spec,
         minibatch_size=args.minibatch_size,
         log_loss_steps=args.log_loss_steps,

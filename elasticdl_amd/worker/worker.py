@@ -37,6 +37,7 @@ class Worker:
         master_client,
         trainer: Trainer,
         data_reader=None,
+        eval_data_reader=None,
         spec: Optional[ModelSpec] = None,
         minibatch_size: int = 32,
         log_loss_steps: int = 100,
@@ -46,6 +47,7 @@ class Worker:
         self.mc = master_client
         self.trainer = trainer
         self.reader = data_reader
+        self.eval_reader = eval_data_reader or data_reader
         self.spec = spec
         self.minibatch_size = minibatch_size
         self.log_loss_steps = log_loss_steps
@@ -57,8 +59,13 @@ class Worker:
 
     # ------------------------------------------------------------- batching
     def _minibatches(self, task: Task) -> Iterator:
+        reader = (
+            self.eval_reader
+            if task.type in (TaskType.EVALUATION, TaskType.PREDICTION)
+            else self.reader
+        )
         records = []
-        for r in self.reader.read_records(task):
+        for r in reader.read_records(task):
             records.append(r)
             if len(records) == self.minibatch_size:
                 yield (self._collate or default_collate)(records)
