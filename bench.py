@@ -61,7 +61,8 @@ def bench_resnet50(args, rank, world, local_rank):
     device = torch.device("cuda", local_rank) if torch.cuda.is_available() else torch.device("cpu")
     if device.type == "cuda":
         torch.cuda.set_device(device)
-        torch.backends.cudnn.benchmark = True  # let MIOpen autotune convs
+        # MIOpen conv autotune (skippable for clean profiling runs)
+        torch.backends.cudnn.benchmark = os.environ.get("EDL_NO_AUTOTUNE") != "1"
 
     torch.manual_seed(1234)
     model = resnet.resnet50(num_classes=args.num_classes)
