@@ -40,6 +40,13 @@ void edl_ht_lookup_or_insert(int64_t*, int32_t*, int64_t, int32_t*, int32_t,
                              int32_t*, hipStream_t);
 void edl_ht_lookup(const int64_t*, const int32_t*, int64_t, const int64_t*,
                    int64_t, int32_t*, hipStream_t);
+void edl_ht_insert_dup(int64_t*, int32_t*, int64_t, int32_t*, int32_t,
+                       const int64_t*, int64_t, int32_t*, int32_t*,
+                       hipStream_t);
+void edl_batch_compact(int32_t*, int32_t*, int64_t, int32_t*, const int32_t*,
+                       int64_t, int32_t*, int32_t*, hipStream_t);
+void edl_accumulate_rows(const float*, const int32_t*, int64_t, int64_t,
+                         float*, hipStream_t);
 void edl_init_new_rows(float*, const int32_t*, const uint8_t*, int64_t,
                        int64_t, uint64_t, float, float, hipStream_t);
 void edl_gather_rows(const float*, const int32_t*, int64_t, int64_t, float*,
@@ -185,6 +192,40 @@ void ht_lookup_or_insert(torch::Tensor keys, torch::Tensor vals,
       cur_stream());
 }
 
+void ht_insert_dup(torch::Tensor keys, torch::Tensor vals,
+                   torch::Tensor row_counter, int64_t max_rows,
+                   torch::Tensor ids, torch::Tensor new_slots,
+                   torch::Tensor error_flag) {
+  TORCH_CHECK((keys.numel() & (keys.numel() - 1)) == 0,
+              "capacity must be a power of two");
+  edl_ht_insert_dup(keys.data_ptr<int64_t>(), vals.data_ptr<int32_t>(),
+                    keys.numel(), row_counter.data_ptr<int32_t>(),
+                    static_cast<int32_t>(max_rows), ids.data_ptr<int64_t>(),
+                    ids.numel(), new_slots.data_ptr<int32_t>(),
+                    error_flag.data_ptr<int32_t>(), cur_stream());
+}
+
+void batch_compact(torch::Tensor ht_keys, torch::Tensor ht_vals,
+                   torch::Tensor counter, torch::Tensor slots,
+                   torch::Tensor unique_slots, torch::Tensor compact_idx) {
+  TORCH_CHECK((ht_keys.numel() & (ht_keys.numel() - 1)) == 0,
+              "capacity must be a power of two");
+  edl_batch_compact(ht_keys.data_ptr<int32_t>(), ht_vals.data_ptr<int32_t>(),
+                    ht_keys.numel(), counter.data_ptr<int32_t>(),
+                    slots.data_ptr<int32_t>(), slots.numel(),
+                    unique_slots.data_ptr<int32_t>(),
+                    compact_idx.data_ptr<int32_t>(), cur_stream());
+}
+
+void accumulate_rows(torch::Tensor grads, torch::Tensor compact_idx,
+                     torch::Tensor acc) {
+  check_f32_cuda(grads, "grads");
+  check_f32_cuda(acc, "acc");
+  edl_accumulate_rows(grads.data_ptr<float>(),
+                      compact_idx.data_ptr<int32_t>(), grads.size(0),
+                      grads.size(1), acc.data_ptr<float>(), cur_stream());
+}
+
 void ht_lookup(torch::Tensor keys, torch::Tensor vals, torch::Tensor ids,
                torch::Tensor out_slots) {
   edl_ht_lookup(keys.data_ptr<int64_t>(), vals.data_ptr<int32_t>(),
@@ -284,6 +325,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sparse_ftrl", &sparse_ftrl);
   m.def("ht_lookup_or_insert", &ht_lookup_or_insert);
   m.def("ht_lookup", &ht_lookup);
+  m.def("ht_insert_dup", &ht_insert_dup);
+  m.def("batch_compact", &batch_compact);
+  m.def("accumulate_rows", &accumulate_rows);
   m.def("init_new_rows", &init_new_rows);
   m.def("gather_rows", &gather_rows);
   m.def("scatter_rows", &scatter_rows);

@@ -409,6 +409,122 @@ __global__ void ht_lookup_or_insert_kernel(
   }
 }
 
+// Duplicate-tolerant insert pass. Lanes holding the same id race on the
+// CAS; exactly one wins, allocates the row and publishes vals[pos];
+// LOSERS DO NOT RESOLVE (no spin — CDNA waves cannot spin on another
+// lane's store). Callers follow with ht_lookup_kernel in a SECOND launch:
+// the kernel boundary makes every winner's vals[] write visible, so the
+// lookup resolves all n (possibly duplicate) ids. This replaces the
+// torch.unique (rocprim sort) pre-pass — ~30 sort/scan kernels per PS
+// step collapse into insert+lookup.
+__global__ void ht_insert_dup_kernel(
+    int64_t* __restrict__ keys, int32_t* __restrict__ vals, int64_t cap_mask,
+    int32_t* __restrict__ row_counter, int32_t max_rows,
+    const int64_t* __restrict__ ids, int64_t n,
+    int32_t* __restrict__ new_slots,   // [n] slot of row created by lane i, else -1
+    int32_t* __restrict__ error_flag) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t id = ids[i];
+    uint64_t h = edl_hash_u64((uint64_t)id) & (uint64_t)cap_mask;
+    int32_t created = -1;
+    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+      int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
+      int64_t cur = keys[pos];
+      if (cur == id) break;  // present (previous launch) or claimed by a peer
+      if (cur == EDL_EMPTY_KEY) {
+        int64_t prev = atomicCAS(reinterpret_cast<unsigned long long*>(&keys[pos]),
+                                 (unsigned long long)EDL_EMPTY_KEY,
+                                 (unsigned long long)id);
+        if (prev == EDL_EMPTY_KEY) {
+          int32_t row = atomicAdd(row_counter, 1);
+          if (row >= max_rows) {
+            atomicExch(error_flag, 1);
+            vals[pos] = 0;
+          } else {
+            vals[pos] = row;
+            created = row;
+          }
+          break;
+        }
+        if (prev == id) break;  // a duplicate lane claimed it
+        // different id won this pos: keep probing
+      }
+    }
+    if (new_slots != nullptr) new_slots[i] = created;
+  }
+}
+
+// Per-batch slot compaction (slot -> dense index) with a scratch hash
+// table keyed by slot. Same duplicate-tolerant two-pass protocol:
+// pass 1 (this kernel) claims; pass 2 (batch_compact_lookup) resolves.
+__global__ void batch_compact_claim_kernel(
+    int32_t* __restrict__ ht_keys,  // [cap] scratch, -1 = empty
+    int32_t* __restrict__ ht_vals, int64_t cap_mask,
+    int32_t* __restrict__ counter,
+    const int32_t* __restrict__ slots, int64_t n,
+    int32_t* __restrict__ unique_slots /* [n] capacity */) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int32_t slot = slots[i];
+    uint64_t h = edl_hash_u64((uint64_t)(uint32_t)slot) & (uint64_t)cap_mask;
+    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+      int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
+      int32_t cur = ht_keys[pos];
+      if (cur == slot) break;
+      if (cur == -1) {
+        int32_t prev = atomicCAS(&ht_keys[pos], -1, slot);
+        if (prev == -1) {
+          int32_t idx = atomicAdd(counter, 1);
+          ht_vals[pos] = idx;
+          unique_slots[idx] = slot;
+          break;
+        }
+        if (prev == slot) break;
+      }
+    }
+  }
+}
+
+__global__ void batch_compact_lookup_kernel(
+    const int32_t* __restrict__ ht_keys, const int32_t* __restrict__ ht_vals,
+    int64_t cap_mask, const int32_t* __restrict__ slots, int64_t n,
+    int32_t* __restrict__ compact_idx) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int32_t slot = slots[i];
+    uint64_t h = edl_hash_u64((uint64_t)(uint32_t)slot) & (uint64_t)cap_mask;
+    int32_t out = 0;
+    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+      int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
+      if (ht_keys[pos] == slot) {
+        out = ht_vals[pos];
+        break;
+      }
+    }
+    compact_idx[i] = out;
+  }
+}
+
+// acc[compact_idx[row], :] += grads[row, :]  (f32 atomic adds; duplicate
+// rows are rare so contention is low)
+__global__ void accumulate_rows_kernel(const float* __restrict__ grads,
+                                       const int32_t* __restrict__ compact_idx,
+                                       int64_t n, int64_t dim,
+                                       float* __restrict__ acc) {
+  int64_t total = n * dim;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int64_t row = i / dim;
+    int64_t col = i - row * dim;
+    atomicAdd(&acc[(int64_t)compact_idx[row] * dim + col], grads[i]);
+  }
+}
+
 __global__ void ht_lookup_kernel(const int64_t* __restrict__ keys,
                                  const int32_t* __restrict__ vals,
                                  int64_t cap_mask,
@@ -515,6 +631,31 @@ void edl_ht_lookup_or_insert(int64_t* keys, int32_t* vals, int64_t capacity,
   ht_lookup_or_insert_kernel<<<grid_for(n), THREADS, 0, s>>>(
       keys, vals, capacity - 1, row_counter, max_rows, ids, n, out_slots,
       out_is_new, error_flag);
+}
+
+void edl_ht_insert_dup(int64_t* keys, int32_t* vals, int64_t capacity,
+                       int32_t* row_counter, int32_t max_rows,
+                       const int64_t* ids, int64_t n, int32_t* new_slots,
+                       int32_t* error_flag, hipStream_t s) {
+  ht_insert_dup_kernel<<<grid_for(n), THREADS, 0, s>>>(
+      keys, vals, capacity - 1, row_counter, max_rows, ids, n, new_slots,
+      error_flag);
+}
+
+void edl_batch_compact(int32_t* ht_keys, int32_t* ht_vals, int64_t capacity,
+                       int32_t* counter, const int32_t* slots, int64_t n,
+                       int32_t* unique_slots, int32_t* compact_idx,
+                       hipStream_t s) {
+  batch_compact_claim_kernel<<<grid_for(n), THREADS, 0, s>>>(
+      ht_keys, ht_vals, capacity - 1, counter, slots, n, unique_slots);
+  batch_compact_lookup_kernel<<<grid_for(n), THREADS, 0, s>>>(
+      ht_keys, ht_vals, capacity - 1, slots, n, compact_idx);
+}
+
+void edl_accumulate_rows(const float* grads, const int32_t* compact_idx,
+                         int64_t n, int64_t dim, float* acc, hipStream_t s) {
+  accumulate_rows_kernel<<<grid_for(n * dim), THREADS, 0, s>>>(
+      grads, compact_idx, n, dim, acc);
 }
 
 void edl_ht_lookup(const int64_t* keys, const int32_t* vals, int64_t capacity,

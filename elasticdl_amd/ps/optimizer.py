@@ -109,9 +109,27 @@ class Optimizer:
 
     def apply_sparse(self, table: EmbeddingTable, grads: torch.Tensor,
                      ids: torch.Tensor, lr_mult: float = 1.0) -> None:
-        """Deduplicate (sum per id), resolve slots, one fused launch."""
-        grads = grads.to(table.device, torch.float32)
-        summed, unique_ids = deduplicate_indexed_slices(grads, ids.to(table.device))
+        """Deduplicate (sum per id), resolve slots, one fused launch.
+
+        GPU path: hash-table compaction + atomic row accumulation (no
+        rocprim sort); CPU path: torch.unique reference."""
+        grads = grads.to(table.device, torch.float32).contiguous()
+        ids = ids.to(table.device)
+        if use_native(table.device):
+            from elasticdl_amd.ops import _C
+
+            slots_full = table.lookup_or_create_dup(ids)
+            unique_slots, compact_idx, u = table.compact_slots(slots_full)
+            if u == slots_full.numel():  # no duplicates: apply directly
+                self._apply_rows(table, grads, slots_full, lr_mult)
+                return
+            acc = torch.zeros(
+                (u, table.dim), dtype=torch.float32, device=table.device
+            )
+            _C.accumulate_rows(grads, compact_idx, acc)
+            self._apply_rows(table, acc, unique_slots.contiguous(), lr_mult)
+            return
+        summed, unique_ids = deduplicate_indexed_slices(grads, ids)
         slots = table.lookup_or_create(unique_ids)
         self._apply_rows(table, summed.contiguous(), slots, lr_mult)
 

@@ -164,6 +164,53 @@ class EmbeddingTable:
         kind, lo, hi = self.initializer
         return (lo, hi)
 
+    def lookup_or_create_dup(self, ids: torch.Tensor) -> torch.Tensor:
+        """Duplicate-tolerant lookup/create: two kernel launches (insert
+        pass + lookup pass) instead of a torch.unique sort pre-pass."""
+        ids = ids.to(self.device, torch.int64).reshape(-1)
+        if not self._native:
+            return self.lookup_or_create(ids)  # CPU dict handles dups
+        n = ids.numel()
+        new_slots = torch.empty(n, dtype=torch.int32, device=self.device)
+        self._C.ht_insert_dup(
+            self._keys, self._vals, self._counter, self.max_rows,
+            ids, new_slots, self._error,
+        )
+        slots = torch.empty(n, dtype=torch.int32, device=self.device)
+        self._C.ht_lookup(self._keys, self._vals, ids, slots)
+        is_new = (new_slots >= 0).to(torch.uint8)
+        lo, hi = self._init_range()
+        self._C.init_new_rows(self.arena, new_slots, is_new, self.seed, lo, hi)
+        mask = is_new.bool()
+        if bool(mask.any()):
+            self._ids_by_slot.index_copy_(
+                0, new_slots[mask].long(), ids[mask]
+            )
+        return slots
+
+    def compact_slots(self, slots: torch.Tensor):
+        """(unique_slots [u], compact_idx [n], u) via the batch scratch
+        hash — GPU only."""
+        assert self._native
+        n = slots.numel()
+        cap = _next_pow2(max(2 * n, 16))
+        if getattr(self, "_bc_cap", 0) < cap:
+            self._bc_keys = torch.empty(cap, dtype=torch.int32, device=self.device)
+            self._bc_vals = torch.empty(cap, dtype=torch.int32, device=self.device)
+            self._bc_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
+            self._bc_cap = cap
+        # reset the scratch (cheap fills; capacity stays the allocated pow2)
+        self._bc_keys.fill_(-1)
+        self._bc_counter.zero_()
+        unique_slots = torch.empty(n, dtype=torch.int32, device=self.device)
+        compact_idx = torch.empty(n, dtype=torch.int32, device=self.device)
+        self._C.batch_compact(
+            self._bc_keys, self._bc_vals, self._bc_counter, slots,
+            unique_slots, compact_idx,
+        )
+        u = int(self._bc_counter.item())
+        return unique_slots[:u], compact_idx, u
+
     def lookup(self, ids: torch.Tensor) -> torch.Tensor:
         """Read-only lookup: slot or -1 per id."""
         ids = ids.to(self.device, torch.int64)
@@ -179,15 +226,19 @@ class EmbeddingTable:
     def gather(self, ids: torch.Tensor, create: bool = True) -> torch.Tensor:
         """Rows for (possibly duplicate) ids; missing rows are created
         (training) or zero (create=False)."""
-        ids = ids.to(self.device, torch.int64)
+        ids = ids.to(self.device, torch.int64).reshape(-1)
+        if self._native:
+            if create:
+                full_slots = self.lookup_or_create_dup(ids)
+            else:
+                full_slots = self.lookup(ids)
+            return self._C.gather_rows(self.arena, full_slots)
         unique_ids, inverse = torch.unique(ids, sorted=True, return_inverse=True)
         if create:
             slots = self.lookup_or_create(unique_ids)
         else:
             slots = self.lookup(unique_ids)
-        full_slots = slots.index_select(0, inverse.view(-1).to(slots.device))
-        if self._native:
-            return self._C.gather_rows(self.arena, full_slots)
+        full_slots = slots.index_select(0, inverse.view(-1))
         return reference.gather_rows(self.arena, full_slots)
 
     # ----------------------------------------------------------- checkpoint

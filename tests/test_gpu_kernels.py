@@ -169,6 +169,74 @@ def test_ht_arena_full_sets_error(C):
     assert int(err.item()) == 1
 
 
+def test_ht_insert_dup_two_pass(C):
+    keys, vals, counter, err, max_rows = _make_ht(C)
+    # heavy duplication: 4096 ids drawn from 100 distinct values
+    base = torch.randperm(1 << 20)[:100].to(torch.int64)
+    ids = base[torch.randint(0, 100, (4096,))].cuda()
+    new_slots = torch.empty(4096, dtype=torch.int32, device="cuda")
+    C.ht_insert_dup(keys, vals, counter, max_rows, ids, new_slots, err)
+    slots = torch.empty(4096, dtype=torch.int32, device="cuda")
+    C.ht_lookup(keys, vals, ids, slots)
+    torch.cuda.synchronize()
+    assert int(err.item()) == 0
+    assert int(counter.item()) == 100
+    assert int((new_slots >= 0).sum().item()) == 100  # one creator per id
+    # same id -> same slot everywhere
+    s = slots.cpu()
+    by_id = {}
+    for i, v in enumerate(ids.cpu().tolist()):
+        if v in by_id:
+            assert by_id[v] == int(s[i])
+        by_id[v] = int(s[i])
+    assert len(by_id) == 100
+
+
+def test_batch_compact_and_accumulate(C):
+    from elasticdl_amd.ps.storage import EmbeddingTable
+
+    t = EmbeddingTable("t", 8, device="cuda", max_rows=1000)
+    ids = torch.tensor([7, 3, 7, 9, 3, 7], dtype=torch.int64)
+    slots = t.lookup_or_create_dup(ids)
+    unique_slots, compact_idx, u = t.compact_slots(slots)
+    torch.cuda.synchronize()
+    assert u == 3
+    # compact_idx maps duplicate rows to the same compact position
+    ci = compact_idx.cpu().tolist()
+    assert ci[0] == ci[2] == ci[5]
+    assert ci[1] == ci[4]
+    grads = torch.ones(6, 8, device="cuda")
+    acc = torch.zeros(u, 8, device="cuda")
+    C.accumulate_rows(grads, compact_idx, acc)
+    torch.cuda.synchronize()
+    sums = sorted(acc.sum(1).div(8).cpu().tolist())
+    assert sums == [1.0, 2.0, 3.0]  # id 9 once, 3 twice, 7 thrice
+
+
+def test_apply_sparse_gpu_dedup_matches_cpu(C):
+    """Duplicate-id sparse Adam: GPU hash-compaction path vs CPU
+    torch.unique reference."""
+    from elasticdl_amd.ps.optimizer import Optimizer
+    from elasticdl_amd.ps.storage import EmbeddingTable
+
+    torch.manual_seed(3)
+    ids = torch.randint(0, 50, (500,), dtype=torch.int64)
+    grads = torch.randn(500, 16)
+
+    results = {}
+    for dev in ("cpu", "cuda"):
+        t = EmbeddingTable("t", 16, device=dev, max_rows=200, seed=42)
+        t.lookup_or_create(torch.arange(50))
+        opt = Optimizer.create("adam", "learning_rate=0.01")
+        opt.begin_apply()
+        opt.apply_sparse(t, grads.to(dev), ids.to(dev))
+        results[dev] = t.gather(torch.arange(50).to(dev)).cpu()
+    torch.cuda.synchronize()
+    assert torch.allclose(results["cpu"], results["cuda"], atol=1e-5), (
+        results["cpu"] - results["cuda"]
+    ).abs().max()
+
+
 # -------------------------- gather / init / scatter ----------------------
 def test_gather_rows(C):
     arena = rand(64, 16).cuda()
