@@ -47,8 +47,9 @@ void edl_batch_compact(int32_t*, int32_t*, int64_t, int32_t*, const int32_t*,
                        int64_t, int32_t*, int32_t*, hipStream_t);
 void edl_accumulate_rows(const float*, const int32_t*, int64_t, int64_t,
                          float*, hipStream_t);
-void edl_init_new_rows(float*, const int32_t*, const uint8_t*, int64_t,
-                       int64_t, uint64_t, float, float, hipStream_t);
+void edl_init_new_rows(float*, const int32_t*, const uint8_t*,
+                       const int64_t*, int64_t, int64_t, uint64_t, float,
+                       float, hipStream_t);
 void edl_gather_rows(const float*, const int32_t*, int64_t, int64_t, float*,
                      hipStream_t);
 void edl_scatter_rows(float*, const int32_t*, const float*, int64_t, int64_t,
@@ -234,10 +235,13 @@ void ht_lookup(torch::Tensor keys, torch::Tensor vals, torch::Tensor ids,
 }
 
 void init_new_rows(torch::Tensor arena, torch::Tensor slots,
-                   torch::Tensor is_new, int64_t seed, double lo, double hi) {
+                   torch::Tensor is_new, torch::Tensor ids, int64_t seed,
+                   double lo, double hi) {
   check_f32_cuda(arena, "arena");
+  TORCH_CHECK(ids.scalar_type() == torch::kInt64);
   edl_init_new_rows(arena.data_ptr<float>(), slots.data_ptr<int32_t>(),
-                    is_new.data_ptr<uint8_t>(), slots.numel(), arena.size(1),
+                    is_new.data_ptr<uint8_t>(), ids.data_ptr<int64_t>(),
+                    slots.numel(), arena.size(1),
                     static_cast<uint64_t>(seed), lo, hi, cur_stream());
 }
 

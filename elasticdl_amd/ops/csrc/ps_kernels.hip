@@ -550,11 +550,14 @@ __global__ void ht_lookup_kernel(const int64_t* __restrict__ keys,
 }
 
 // Lazy row init: uniform RNG in [lo, hi) from a stateless splitmix hash of
-// (seed, slot, col) — rows are reproducible given the creation seed
-// (reference inits uniform(-0.05, 0.05): embedding_table.go:40-58).
+// (seed, ID, col) — keyed on the embedding ID, not the arena slot, so the
+// value of a row is deterministic regardless of the (atomics-ordered,
+// nondeterministic) slot assignment. Reproduces the reference's
+// init-on-first-touch (embedding_table.go:40-58) deterministically.
 __global__ void init_new_rows_kernel(float* __restrict__ arena,
                                      const int32_t* __restrict__ slots,
                                      const uint8_t* __restrict__ is_new,
+                                     const int64_t* __restrict__ ids,
                                      int64_t n, int64_t dim, uint64_t seed,
                                      float lo, float hi) {
   int64_t total = n * dim;
@@ -565,7 +568,8 @@ __global__ void init_new_rows_kernel(float* __restrict__ arena,
     if (!is_new[row]) continue;
     int64_t col = i - row * dim;
     int64_t slot = slots[row];
-    uint64_t r = edl_hash_u64(seed ^ ((uint64_t)slot << 32) ^ (uint64_t)col);
+    uint64_t x = edl_hash_u64(seed ^ (uint64_t)ids[row]);
+    uint64_t r = edl_hash_u64(x ^ (uint64_t)col);
     float u = (float)(r >> 40) * (1.0f / 16777216.0f);  // [0,1) from top 24 bits
     arena[slot * dim + col] = lo + u * (hi - lo);
   }
@@ -666,10 +670,11 @@ void edl_ht_lookup(const int64_t* keys, const int32_t* vals, int64_t capacity,
 }
 
 void edl_init_new_rows(float* arena, const int32_t* slots,
-                       const uint8_t* is_new, int64_t n, int64_t dim,
-                       uint64_t seed, float lo, float hi, hipStream_t s) {
+                       const uint8_t* is_new, const int64_t* ids, int64_t n,
+                       int64_t dim, uint64_t seed, float lo, float hi,
+                       hipStream_t s) {
   init_new_rows_kernel<<<grid_for(n * dim), THREADS, 0, s>>>(
-      arena, slots, is_new, n, dim, seed, lo, hi);
+      arena, slots, is_new, ids, n, dim, seed, lo, hi);
 }
 
 void edl_gather_rows(const float* arena, const int32_t* slots, int64_t n,

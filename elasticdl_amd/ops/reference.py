@@ -126,17 +126,11 @@ def _splitmix64(x: torch.Tensor) -> torch.Tensor:
     return x ^ (x >> 31)
 
 
-def init_rows_values(slots: torch.Tensor, dim: int, seed: int, lo: float, hi: float):
-    """Deterministic uniform rows, same bits as init_new_rows_kernel."""
-    s = slots.to(torch.int64).view(-1, 1)
-    cols = torch.arange(dim, dtype=torch.int64, device=slots.device).view(1, -1)
-    # uint64 math emulated in int64: wrap-around is identical mod 2^64,
-    # but python ints avoid overflow issues -> do it on CPU via numpy uint64
+def init_rows_values(ids: torch.Tensor, dim: int, seed: int, lo: float, hi: float):
+    """Deterministic uniform rows, same bits as init_new_rows_kernel:
+    r = splitmix64(splitmix64(seed ^ id) ^ col), keyed on the embedding ID
+    so values are independent of arena slot assignment order."""
     import numpy as np
-
-    x = (np.uint64(seed) ^ (s.cpu().numpy().astype(np.uint64) << np.uint64(32))
-         ^ cols.cpu().numpy().astype(np.uint64))
-    x = x.astype(np.uint64)
 
     def sm64(v):
         v = v + np.uint64(0x9E3779B97F4A7C15)
@@ -144,11 +138,14 @@ def init_rows_values(slots: torch.Tensor, dim: int, seed: int, lo: float, hi: fl
         v = (v ^ (v >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
         return v ^ (v >> np.uint64(31))
 
+    idv = ids.detach().cpu().to(torch.int64).view(-1, 1).numpy().astype(np.uint64)
+    cols = np.arange(dim, dtype=np.uint64).reshape(1, -1)
     with np.errstate(over="ignore"):
-        r = sm64(x)
+        x = sm64(np.uint64(seed) ^ idv)
+        r = sm64(x ^ cols)
     u = (r >> np.uint64(40)).astype(np.float32) * np.float32(1.0 / 16777216.0)
     vals = lo + u * (hi - lo)
-    return torch.from_numpy(vals).to(slots.device)
+    return torch.from_numpy(vals).to(ids.device)
 
 
 def gather_rows(arena: torch.Tensor, slots: torch.Tensor) -> torch.Tensor:

@@ -128,7 +128,9 @@ class EmbeddingTable:
                 self._error,
             )
             lo, hi = self._init_range()
-            self._C.init_new_rows(self.arena, slots, is_new, self.seed, lo, hi)
+            self._C.init_new_rows(
+                self.arena, slots, is_new, unique_ids, self.seed, lo, hi
+            )
             # remember id per slot for checkpoint export
             new_mask = is_new.bool()
             if bool(new_mask.any()):
@@ -148,13 +150,14 @@ class EmbeddingTable:
                 self._id_to_slot[v] = s
                 self._n_rows += 1
                 self._ensure_capacity(self._n_rows)
-                new_slots.append(s)
+                new_slots.append((s, v))
             slots[i] = s
         if new_slots:
-            ns = torch.tensor(new_slots, dtype=torch.int32)
+            ns = torch.tensor([s for s, _ in new_slots], dtype=torch.int32)
+            nids = torch.tensor([v for _, v in new_slots], dtype=torch.int64)
             lo, hi = self._init_range()
             self.arena[ns.long()] = reference.init_rows_values(
-                ns, self.dim, self.seed, lo, hi
+                nids, self.dim, self.seed, lo, hi
             )
         return slots
 
@@ -180,7 +183,9 @@ class EmbeddingTable:
         self._C.ht_lookup(self._keys, self._vals, ids, slots)
         is_new = (new_slots >= 0).to(torch.uint8)
         lo, hi = self._init_range()
-        self._C.init_new_rows(self.arena, new_slots, is_new, self.seed, lo, hi)
+        self._C.init_new_rows(
+            self.arena, new_slots, is_new, ids, self.seed, lo, hi
+        )
         mask = is_new.bool()
         if bool(mask.any()):
             self._ids_by_slot.index_copy_(

@@ -251,10 +251,11 @@ def test_init_rows_bit_identical_to_cpu(C):
     dim, seed = 8, 12345
     arena = torch.zeros(32, dim, device="cuda")
     slots = torch.arange(10, dtype=torch.int32, device="cuda")
+    ids = (torch.arange(10, dtype=torch.int64) * 977 + 13).cuda()
     is_new = torch.ones(10, dtype=torch.uint8, device="cuda")
-    C.init_new_rows(arena, slots, is_new, seed, -0.05, 0.05)
+    C.init_new_rows(arena, slots, is_new, ids, seed, -0.05, 0.05)
     torch.cuda.synchronize()
-    ref = reference.init_rows_values(slots.cpu(), dim, seed, -0.05, 0.05)
+    ref = reference.init_rows_values(ids.cpu(), dim, seed, -0.05, 0.05)
     assert torch.allclose(arena[:10].cpu(), ref, atol=0), (
         arena[:10].cpu() - ref
     ).abs().max()
