@@ -8,4 +8,7 @@ from elasticdl_amd.preprocessing.layers import (  # noqa: F401
     RoundIdentity,
     SparseEmbedding,
     ToNumber,
+    fit_normalizer,
+    to_padded,
+    to_sparse,
 )

@@ -222,6 +222,34 @@ class ConcatenateWithOffset(nn.Module):
         return torch.cat(shifted, dim=1)
 
 
+def to_padded(values: Sequence[Sequence], pad_value: int = PAD,
+              dtype=torch.int64) -> torch.Tensor:
+    """List-of-lists -> [batch, max_len] padded tensor — the torch-native
+    stand-in for the reference's ToRagged layer (to_ragged.py); all
+    variable-length layers here consume this convention."""
+    max_len = max((len(r) for r in values), default=0)
+    out = torch.full((len(values), max(max_len, 1)), pad_value, dtype=dtype)
+    for i, row in enumerate(values):
+        if len(row):
+            out[i, :len(row)] = torch.as_tensor(list(row), dtype=dtype)
+    return out
+
+
+def to_sparse(padded: torch.Tensor, pad_value: int = PAD):
+    """Padded tensor -> torch sparse COO (reference: to_sparse.py)."""
+    mask = padded != pad_value
+    idx = mask.nonzero().t()
+    return torch.sparse_coo_tensor(idx, padded[mask], padded.shape)
+
+
+def fit_normalizer(values: torch.Tensor) -> "Normalizer":
+    """Build a Normalizer from data statistics — the offline-analysis
+    stand-in for the reference's analyzer_utils (SQL-computed stats)."""
+    v = values.float()
+    std = float(v.std())
+    return Normalizer(subtractor=float(v.mean()), divisor=std if std > 0 else 1.0)
+
+
 class SparseEmbedding(nn.Module):
     """Local (non-PS) embedding over padded variable-length ids with a
     combiner (reference: sparse_embedding.py:20). For the PS-distributed

@@ -19,6 +19,9 @@ _BUILTIN = {
     "resnet50": "elasticdl_amd.models.resnet",
     "wide_deep": "elasticdl_amd.models.wide_deep",
     "deepfm": "elasticdl_amd.models.deepfm",
+    "dcn": "elasticdl_amd.models.dcn",
+    "cifar10": "elasticdl_amd.models.cifar10",
+    "iris": "elasticdl_amd.models.iris",
 }
 
 

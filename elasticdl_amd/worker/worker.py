@@ -53,7 +53,9 @@ class Worker:
         self.log_loss_steps = log_loss_steps
         self.export_path = export_path
         self.shard_service = DataShardService(master_client, minibatch_size)
-        self.timing = Timing(enabled=False)
+        import os
+
+        self.timing = Timing(enabled=os.environ.get("EDL_TIMING") == "1")
         self._collate = getattr(spec.module, "collate_fn", None) if spec else None
         self._step = 0
 
@@ -99,11 +101,18 @@ class Worker:
 
     def _run_training_task(self, task: Task) -> None:
         n_records = 0
+        self.timing.start_record_time("task_process")
         for batch in self._minibatches(task):
             batch_records = _batch_len(batch)
+            self.timing.start_record_time("batch_process")
             self._process_minibatch(batch, train=True)
+            self.timing.end_record_time("batch_process")
             n_records += batch_records
             self.shard_service.report_batch_done(batch_records)
+        self.timing.end_record_time("task_process")
+        if self.timing.enabled:
+            logger.info("Task %d timing: %s", task.task_id,
+                        self.timing.report_timing(reset=True))
 
     def _run_evaluation_task(self, task: Task) -> None:
         outputs, labels = [], []
