@@ -42,15 +42,27 @@ def _all_to_all_tensor(
     send = [t.to(device, dtype).contiguous() for t in send]
     if backend == "nccl":
         counts_d = counts.to(device)
-        recv_counts = torch.zeros_like(counts_d)
-        dist.all_to_all_single(recv_counts, counts_d, group=group)
-        recv_counts = recv_counts.cpu().tolist()
-        recv = [
-            torch.empty((n, *trailing), dtype=dtype, device=device)
-            for n in recv_counts
-        ]
-        dist.all_to_all(recv, send, group=group)
-        return recv
+        recv_counts_d = torch.zeros_like(counts_d)
+        dist.all_to_all_single(recv_counts_d, counts_d, group=group)
+        recv_counts = recv_counts_d.cpu().tolist()
+        send_counts = counts.tolist()
+        # single fused exchange with explicit splits — zero-size splits are
+        # well-defined (dist.all_to_all with empty member tensors is not)
+        flat_in = (
+            torch.cat(send, dim=0)
+            if sum(send_counts)
+            else torch.empty((0, *trailing), dtype=dtype, device=device)
+        )
+        flat_out = torch.empty(
+            (sum(recv_counts), *trailing), dtype=dtype, device=device
+        )
+        dist.all_to_all_single(
+            flat_out, flat_in,
+            output_split_sizes=recv_counts,
+            input_split_sizes=send_counts,
+            group=group,
+        )
+        return list(torch.split(flat_out, recv_counts, dim=0))
     # ---- gloo emulation
     gathered = [torch.zeros_like(counts) for _ in range(world)]
     dist.all_gather(gathered, counts, group=group)
