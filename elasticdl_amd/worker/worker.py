@@ -60,7 +60,7 @@ class Worker:
         self._step = 0
 
     # ------------------------------------------------------------- batching
-    def _minibatches(self, task: Task) -> Iterator:
+    def _raw_minibatches(self, task: Task) -> Iterator:
         reader = (
             self.eval_reader
             if task.type in (TaskType.EVALUATION, TaskType.PREDICTION)
@@ -74,6 +74,34 @@ class Worker:
                 records = []
         if records:
             yield (self._collate or default_collate)(records)
+
+    def _minibatches(self, task: Task, prefetch: int = 2) -> Iterator:
+        """Background-thread prefetch so record IO + collate overlap with
+        the training step (reference: dataset.prefetch(1), worker.py:334)."""
+        import queue
+        import threading
+
+        q: "queue.Queue" = queue.Queue(maxsize=prefetch)
+        _END = object()
+
+        def producer():
+            try:
+                for batch in self._raw_minibatches(task):
+                    q.put(batch)
+                q.put(_END)
+            except BaseException as e:  # noqa: BLE001 - propagate to consumer
+                q.put(e)
+
+        t = threading.Thread(target=producer, daemon=True)
+        t.start()
+        while True:
+            item = q.get()
+            if item is _END:
+                break
+            if isinstance(item, BaseException):
+                raise item
+            yield item
+        t.join(5)
 
     # ------------------------------------------------------------- training
     def _process_minibatch(self, batch, train: bool):
