@@ -16,7 +16,7 @@ and the driver's multi-GPU scaling runs.
 import datetime
 import os
 import time
-from typing import Callable, Optional
+from typing import Optional
 
 import torch
 import torch.distributed as dist

@@ -9,7 +9,7 @@ HIP segmented-sum kernel in elasticdl_amd.ops.
 """
 
 from dataclasses import dataclass
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 

@@ -12,7 +12,7 @@ Also provides the sparse/bag input form with sum/mean/sqrtn combiners
 (reference: embedding.py:117-132, safe_embedding_lookup_sparse).
 """
 
-from typing import Callable, Dict, List, Optional
+from typing import Callable, List, Optional
 
 import torch
 import torch.nn as nn

@@ -9,7 +9,6 @@ pod_manager.py:577-604.
 
 import os
 import subprocess
-import sys
 import threading
 import time
 from typing import Callable, Dict, List, Optional

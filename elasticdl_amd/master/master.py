@@ -15,7 +15,7 @@ import time
 from typing import List, Optional
 
 from elasticdl_amd.common.args import parse_model_params
-from elasticdl_amd.common.constants import DistributionStrategy, WorkerEnv
+from elasticdl_amd.common.constants import DistributionStrategy
 from elasticdl_amd.common.log_utils import default_logger as logger
 from elasticdl_amd.common.rpc import start_server
 from elasticdl_amd.master.evaluation_service import EvaluationService

@@ -17,7 +17,7 @@ shrink the world) — reference servicer.py:111-125.
 """
 
 import threading
-from typing import Callable, Dict, Optional
+from typing import Callable, Dict
 
 from elasticdl_amd.common.log_utils import default_logger as logger
 from elasticdl_amd.common.task import Task, TaskType

@@ -14,7 +14,7 @@ on tf dense/sparse/ragged tensors; the torch-native convention here is:
 
 import hashlib
 import math
-from typing import Dict, List, Optional, Sequence, Union
+from typing import List, Optional, Sequence
 
 import torch
 import torch.nn as nn

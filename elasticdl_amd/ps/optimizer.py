@@ -16,7 +16,7 @@ opt_args string format matches the Go PS CLI contract
 (optimizer.go:284-390): "learning_rate=0.1;momentum=0.9;nesterov=false".
 """
 
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 

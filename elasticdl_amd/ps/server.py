@@ -12,7 +12,6 @@ push_gradients) on the raw-bytes codec, backed by the GPU engine. Extras:
 import argparse
 import os
 import threading
-import time
 from typing import Optional
 
 import torch

@@ -16,7 +16,7 @@ sharded across 8 GPUs' HBM3E). Works on gloo/CPU for tests (world>1) and
 degrades to the purely local engine at world=1.
 """
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 import torch.distributed as dist

@@ -20,7 +20,7 @@ On CPU (tests, local mode) the same API is served by a python dict +
 growable arena with bit-identical RNG init.
 """
 
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 

@@ -8,7 +8,7 @@ materializes trained PS rows back into stock ``nn.Embedding`` layers for
 export.
 """
 
-from typing import Dict, List, Tuple
+from typing import List, Tuple
 
 import torch
 import torch.nn as nn

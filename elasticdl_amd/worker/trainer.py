@@ -1,7 +1,7 @@
 """Trainer interface (reference: worker/trainer.py:17-54) + the Local
 trainer used by local mode and tests."""
 
-from typing import Dict, Optional, Tuple
+from typing import Tuple
 
 import torch
 
