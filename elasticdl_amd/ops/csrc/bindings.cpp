@@ -37,11 +37,13 @@ void edl_sparse_ftrl(float*, float*, float*, const float*, const int32_t*,
                      hipStream_t);
 void edl_ht_lookup_or_insert(int64_t*, int32_t*, int64_t, int32_t*, int32_t,
                              const int64_t*, int64_t, int32_t*, uint8_t*,
-                             int32_t*, hipStream_t);
+                             int32_t*, int64_t*, hipStream_t);
+void edl_detect_dup_slots(const int32_t*, int64_t, int32_t*, int32_t,
+                          int32_t*, hipStream_t);
 void edl_ht_lookup(const int64_t*, const int32_t*, int64_t, const int64_t*,
                    int64_t, int32_t*, hipStream_t);
 void edl_ht_insert_dup(int64_t*, int32_t*, int64_t, int32_t*, int32_t,
-                       const int64_t*, int64_t, int32_t*, int32_t*,
+                       const int64_t*, int64_t, int32_t*, int32_t*, int64_t*,
                        hipStream_t);
 void edl_batch_compact(int32_t*, int32_t*, int64_t, int32_t*, const int32_t*,
                        int64_t, int32_t*, int32_t*, hipStream_t);
@@ -49,7 +51,7 @@ void edl_accumulate_rows(const float*, const int32_t*, int64_t, int64_t,
                          float*, hipStream_t);
 void edl_init_new_rows(float*, const int32_t*, const uint8_t*,
                        const int64_t*, int64_t, int64_t, uint64_t, float,
-                       float, hipStream_t);
+                       float, hipStream_t);  // is_new may be null: slot<0 skips
 void edl_gather_rows(const float*, const int32_t*, int64_t, int64_t, float*,
                      hipStream_t);
 void edl_scatter_rows(float*, const int32_t*, const float*, int64_t, int64_t,
@@ -180,7 +182,8 @@ void sparse_ftrl(torch::Tensor arena, torch::Tensor z, torch::Tensor n,
 void ht_lookup_or_insert(torch::Tensor keys, torch::Tensor vals,
                          torch::Tensor row_counter, int64_t max_rows,
                          torch::Tensor ids, torch::Tensor out_slots,
-                         torch::Tensor out_is_new, torch::Tensor error_flag) {
+                         torch::Tensor out_is_new, torch::Tensor error_flag,
+                         c10::optional<torch::Tensor> ids_by_slot) {
   TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
   TORCH_CHECK((keys.numel() & (keys.numel() - 1)) == 0,
               "capacity must be a power of two");
@@ -190,20 +193,32 @@ void ht_lookup_or_insert(torch::Tensor keys, torch::Tensor vals,
       row_counter.data_ptr<int32_t>(), static_cast<int32_t>(max_rows),
       ids.data_ptr<int64_t>(), ids.numel(), out_slots.data_ptr<int32_t>(),
       out_is_new.data_ptr<uint8_t>(), error_flag.data_ptr<int32_t>(),
+      ids_by_slot.has_value() ? ids_by_slot->data_ptr<int64_t>() : nullptr,
       cur_stream());
+}
+
+void detect_dup_slots(torch::Tensor slots, torch::Tensor mark, int64_t tag,
+                      torch::Tensor dup_flag) {
+  edl_detect_dup_slots(slots.data_ptr<int32_t>(), slots.numel(),
+                       mark.data_ptr<int32_t>(), static_cast<int32_t>(tag),
+                       dup_flag.data_ptr<int32_t>(), cur_stream());
 }
 
 void ht_insert_dup(torch::Tensor keys, torch::Tensor vals,
                    torch::Tensor row_counter, int64_t max_rows,
                    torch::Tensor ids, torch::Tensor new_slots,
-                   torch::Tensor error_flag) {
+                   torch::Tensor error_flag,
+                   c10::optional<torch::Tensor> ids_by_slot) {
   TORCH_CHECK((keys.numel() & (keys.numel() - 1)) == 0,
               "capacity must be a power of two");
   edl_ht_insert_dup(keys.data_ptr<int64_t>(), vals.data_ptr<int32_t>(),
                     keys.numel(), row_counter.data_ptr<int32_t>(),
                     static_cast<int32_t>(max_rows), ids.data_ptr<int64_t>(),
                     ids.numel(), new_slots.data_ptr<int32_t>(),
-                    error_flag.data_ptr<int32_t>(), cur_stream());
+                    error_flag.data_ptr<int32_t>(),
+                    ids_by_slot.has_value() ? ids_by_slot->data_ptr<int64_t>()
+                                            : nullptr,
+                    cur_stream());
 }
 
 void batch_compact(torch::Tensor ht_keys, torch::Tensor ht_vals,
@@ -235,13 +250,13 @@ void ht_lookup(torch::Tensor keys, torch::Tensor vals, torch::Tensor ids,
 }
 
 void init_new_rows(torch::Tensor arena, torch::Tensor slots,
-                   torch::Tensor is_new, torch::Tensor ids, int64_t seed,
-                   double lo, double hi) {
+                   c10::optional<torch::Tensor> is_new, torch::Tensor ids,
+                   int64_t seed, double lo, double hi) {
   check_f32_cuda(arena, "arena");
   TORCH_CHECK(ids.scalar_type() == torch::kInt64);
   edl_init_new_rows(arena.data_ptr<float>(), slots.data_ptr<int32_t>(),
-                    is_new.data_ptr<uint8_t>(), ids.data_ptr<int64_t>(),
-                    slots.numel(), arena.size(1),
+                    is_new.has_value() ? is_new->data_ptr<uint8_t>() : nullptr,
+                    ids.data_ptr<int64_t>(), slots.numel(), arena.size(1),
                     static_cast<uint64_t>(seed), lo, hi, cur_stream());
 }
 
@@ -330,6 +345,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ht_lookup_or_insert", &ht_lookup_or_insert);
   m.def("ht_lookup", &ht_lookup);
   m.def("ht_insert_dup", &ht_insert_dup);
+  m.def("detect_dup_slots", &detect_dup_slots);
   m.def("batch_compact", &batch_compact);
   m.def("accumulate_rows", &accumulate_rows);
   m.def("init_new_rows", &init_new_rows);

@@ -363,7 +363,8 @@ __global__ void ht_lookup_or_insert_kernel(
     int64_t* __restrict__ keys, int32_t* __restrict__ vals, int64_t cap_mask,
     int32_t* __restrict__ row_counter, int32_t max_rows,
     const int64_t* __restrict__ ids, int64_t n, int32_t* __restrict__ out_slots,
-    uint8_t* __restrict__ out_is_new, int32_t* __restrict__ error_flag) {
+    uint8_t* __restrict__ out_is_new, int32_t* __restrict__ error_flag,
+    int64_t* __restrict__ ids_by_slot) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -392,6 +393,7 @@ __global__ void ht_lookup_or_insert_kernel(
             break;
           }
           vals[pos] = row;
+          if (ids_by_slot != nullptr) ids_by_slot[row] = id;
           slot = row;
           is_new = 1;
           break;
@@ -422,7 +424,8 @@ __global__ void ht_insert_dup_kernel(
     int32_t* __restrict__ row_counter, int32_t max_rows,
     const int64_t* __restrict__ ids, int64_t n,
     int32_t* __restrict__ new_slots,   // [n] slot of row created by lane i, else -1
-    int32_t* __restrict__ error_flag) {
+    int32_t* __restrict__ error_flag,
+    int64_t* __restrict__ ids_by_slot) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -444,6 +447,7 @@ __global__ void ht_insert_dup_kernel(
             vals[pos] = 0;
           } else {
             vals[pos] = row;
+            if (ids_by_slot != nullptr) ids_by_slot[row] = id;
             created = row;
           }
           break;
@@ -509,6 +513,22 @@ __global__ void batch_compact_lookup_kernel(
   }
 }
 
+// Epoch-tagged duplicate detection: mark[slot] stores the last batch tag
+// that touched it; a lane seeing its own tag already present flags a
+// duplicate. Scattered atomicExch (no shared counter) -> no contention;
+// the mark array needs no reset between batches (tag increases).
+__global__ void detect_dup_slots_kernel(const int32_t* __restrict__ slots,
+                                        int64_t n, int32_t* __restrict__ mark,
+                                        int32_t tag,
+                                        int32_t* __restrict__ dup_flag) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int32_t old = atomicExch(&mark[slots[i]], tag);
+    if (old == tag) atomicExch(dup_flag, 1);
+  }
+}
+
 // acc[compact_idx[row], :] += grads[row, :]  (f32 atomic adds; duplicate
 // rows are rare so contention is low)
 __global__ void accumulate_rows_kernel(const float* __restrict__ grads,
@@ -565,7 +585,7 @@ __global__ void init_new_rows_kernel(float* __restrict__ arena,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
     int64_t row = i / dim;
-    if (!is_new[row]) continue;
+    if (is_new != nullptr ? !is_new[row] : slots[row] < 0) continue;
     int64_t col = i - row * dim;
     int64_t slot = slots[row];
     uint64_t x = edl_hash_u64(seed ^ (uint64_t)ids[row]);
@@ -631,19 +651,26 @@ void edl_ht_lookup_or_insert(int64_t* keys, int32_t* vals, int64_t capacity,
                              int32_t* row_counter, int32_t max_rows,
                              const int64_t* ids, int64_t n, int32_t* out_slots,
                              uint8_t* out_is_new, int32_t* error_flag,
-                             hipStream_t s) {
+                             int64_t* ids_by_slot, hipStream_t s) {
   ht_lookup_or_insert_kernel<<<grid_for(n), THREADS, 0, s>>>(
       keys, vals, capacity - 1, row_counter, max_rows, ids, n, out_slots,
-      out_is_new, error_flag);
+      out_is_new, error_flag, ids_by_slot);
 }
 
 void edl_ht_insert_dup(int64_t* keys, int32_t* vals, int64_t capacity,
                        int32_t* row_counter, int32_t max_rows,
                        const int64_t* ids, int64_t n, int32_t* new_slots,
-                       int32_t* error_flag, hipStream_t s) {
+                       int32_t* error_flag, int64_t* ids_by_slot,
+                       hipStream_t s) {
   ht_insert_dup_kernel<<<grid_for(n), THREADS, 0, s>>>(
       keys, vals, capacity - 1, row_counter, max_rows, ids, n, new_slots,
-      error_flag);
+      error_flag, ids_by_slot);
+}
+
+void edl_detect_dup_slots(const int32_t* slots, int64_t n, int32_t* mark,
+                          int32_t tag, int32_t* dup_flag, hipStream_t s) {
+  detect_dup_slots_kernel<<<grid_for(n), THREADS, 0, s>>>(slots, n, mark, tag,
+                                                          dup_flag);
 }
 
 void edl_batch_compact(int32_t* ht_keys, int32_t* ht_vals, int64_t capacity,

@@ -119,10 +119,11 @@ class Optimizer:
             from elasticdl_amd.ops import _C
 
             slots_full = table.lookup_or_create_dup(ids)
-            unique_slots, compact_idx, u = table.compact_slots(slots_full)
-            if u == slots_full.numel():  # no duplicates: apply directly
+            # common case: no in-batch duplicates -> apply directly
+            if not table.has_duplicate_slots(slots_full):
                 self._apply_rows(table, grads, slots_full, lr_mult)
                 return
+            unique_slots, compact_idx, u = table.compact_slots(slots_full)
             acc = torch.zeros(
                 (u, table.dim), dtype=torch.float32, device=table.device
             )

@@ -126,17 +126,12 @@ class EmbeddingTable:
                 slots,
                 is_new,
                 self._error,
+                self._ids_by_slot,  # id bookkeeping done in-kernel
             )
             lo, hi = self._init_range()
             self._C.init_new_rows(
                 self.arena, slots, is_new, unique_ids, self.seed, lo, hi
             )
-            # remember id per slot for checkpoint export
-            new_mask = is_new.bool()
-            if bool(new_mask.any()):
-                self._ids_by_slot.index_copy_(
-                    0, slots[new_mask].long(), unique_ids[new_mask]
-                )
             return slots
         # ----- CPU path
         slots = torch.empty(unique_ids.numel(), dtype=torch.int32)
@@ -177,21 +172,33 @@ class EmbeddingTable:
         new_slots = torch.empty(n, dtype=torch.int32, device=self.device)
         self._C.ht_insert_dup(
             self._keys, self._vals, self._counter, self.max_rows,
-            ids, new_slots, self._error,
+            ids, new_slots, self._error, self._ids_by_slot,
         )
         slots = torch.empty(n, dtype=torch.int32, device=self.device)
         self._C.ht_lookup(self._keys, self._vals, ids, slots)
-        is_new = (new_slots >= 0).to(torch.uint8)
         lo, hi = self._init_range()
+        # is_new=None: rows with new_slots < 0 are skipped in-kernel
         self._C.init_new_rows(
-            self.arena, new_slots, is_new, ids, self.seed, lo, hi
+            self.arena, new_slots, None, ids, self.seed, lo, hi
         )
-        mask = is_new.bool()
-        if bool(mask.any()):
-            self._ids_by_slot.index_copy_(
-                0, new_slots[mask].long(), ids[mask]
-            )
         return slots
+
+    def has_duplicate_slots(self, slots: torch.Tensor) -> bool:
+        """Epoch-tagged scatter check (no sort, no shared counter); one
+        device->host flag read."""
+        assert self._native
+        if getattr(self, "_mark", None) is None:
+            self._mark = torch.zeros(
+                self.max_rows, dtype=torch.int32, device=self.device
+            )
+            self._mark_tag = 0
+            self._dup_flag = torch.zeros(
+                1, dtype=torch.int32, device=self.device
+            )
+        self._mark_tag += 1
+        self._dup_flag.zero_()
+        self._C.detect_dup_slots(slots, self._mark, self._mark_tag, self._dup_flag)
+        return bool(self._dup_flag.item())
 
     def compact_slots(self, slots: torch.Tensor):
         """(unique_slots [u], compact_idx [n], u) via the batch scratch

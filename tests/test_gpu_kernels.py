@@ -134,7 +134,7 @@ def test_ht_insert_lookup_bulk(C):
     ids = torch.randperm(1 << 20)[: 10000].to(torch.int64).cuda()
     slots = torch.empty(10000, dtype=torch.int32, device="cuda")
     is_new = torch.empty(10000, dtype=torch.uint8, device="cuda")
-    C.ht_lookup_or_insert(keys, vals, counter, max_rows, ids, slots, is_new, err)
+    C.ht_lookup_or_insert(keys, vals, counter, max_rows, ids, slots, is_new, err, None)
     torch.cuda.synchronize()
     assert int(err.item()) == 0
     assert int(counter.item()) == 10000
@@ -145,7 +145,7 @@ def test_ht_insert_lookup_bulk(C):
     assert s.unique().numel() == 10000
     # second call: same slots, nothing new
     slots2 = torch.empty_like(slots)
-    C.ht_lookup_or_insert(keys, vals, counter, max_rows, ids, slots2, is_new, err)
+    C.ht_lookup_or_insert(keys, vals, counter, max_rows, ids, slots2, is_new, err, None)
     torch.cuda.synchronize()
     assert torch.equal(slots, slots2)
     assert not bool(is_new.any())
@@ -164,7 +164,7 @@ def test_ht_arena_full_sets_error(C):
     ids = torch.arange(20, dtype=torch.int64).cuda()
     slots = torch.empty(20, dtype=torch.int32, device="cuda")
     is_new = torch.empty(20, dtype=torch.uint8, device="cuda")
-    C.ht_lookup_or_insert(keys, vals, counter, 10, ids, slots, is_new, err)
+    C.ht_lookup_or_insert(keys, vals, counter, 10, ids, slots, is_new, err, None)
     torch.cuda.synchronize()
     assert int(err.item()) == 1
 
@@ -175,7 +175,7 @@ def test_ht_insert_dup_two_pass(C):
     base = torch.randperm(1 << 20)[:100].to(torch.int64)
     ids = base[torch.randint(0, 100, (4096,))].cuda()
     new_slots = torch.empty(4096, dtype=torch.int32, device="cuda")
-    C.ht_insert_dup(keys, vals, counter, max_rows, ids, new_slots, err)
+    C.ht_insert_dup(keys, vals, counter, max_rows, ids, new_slots, err, None)
     slots = torch.empty(4096, dtype=torch.int32, device="cuda")
     C.ht_lookup(keys, vals, ids, slots)
     torch.cuda.synchronize()
@@ -190,6 +190,18 @@ def test_ht_insert_dup_two_pass(C):
             assert by_id[v] == int(s[i])
         by_id[v] = int(s[i])
     assert len(by_id) == 100
+
+
+def test_detect_dup_slots(C):
+    from elasticdl_amd.ps.storage import EmbeddingTable
+
+    t = EmbeddingTable("t", 8, device="cuda", max_rows=1000)
+    unique = t.lookup_or_create_dup(torch.tensor([1, 2, 3], dtype=torch.int64))
+    assert not t.has_duplicate_slots(unique)
+    dup = t.lookup_or_create_dup(torch.tensor([1, 2, 1], dtype=torch.int64))
+    assert t.has_duplicate_slots(dup)
+    # tags advance: a fresh unique batch after a dup batch is clean
+    assert not t.has_duplicate_slots(unique)
 
 
 def test_batch_compact_and_accumulate(C):
