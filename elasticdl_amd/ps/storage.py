@@ -73,10 +73,16 @@ class EmbeddingTable:
                 (max_rows, dim), dtype=torch.float32, device=self.device
             )
         else:
+            import threading
+
             self._id_to_slot: Dict[int, int] = {}
             self._grow = max(256, min(max_rows, 65536))
             self.arena = torch.empty((self._grow, dim), dtype=torch.float32)
             self._n_rows = 0
+            # CPU path: dict check-then-insert must be atomic under the
+            # PS server's thread pool (GPU path is serialized by the HIP
+            # stream and CAS-protected in the hash table)
+            self._cpu_lock = threading.Lock()
 
     # ----------------------------------------------------------- properties
     @property
@@ -134,6 +140,10 @@ class EmbeddingTable:
             )
             return slots
         # ----- CPU path
+        with self._cpu_lock:
+            return self._cpu_lookup_or_create(unique_ids)
+
+    def _cpu_lookup_or_create(self, unique_ids: torch.Tensor) -> torch.Tensor:
         slots = torch.empty(unique_ids.numel(), dtype=torch.int32)
         new_slots = []
         for i, v in enumerate(unique_ids.tolist()):
