@@ -24,6 +24,20 @@ import torch.distributed as dist
 from elasticdl_amd.common.log_utils import default_logger as logger
 
 
+class CollectiveFailureError(RuntimeError):
+    """Raised when the elastic re-init retry budget is exhausted; worker
+    task loops must treat this as a task failure, NOT as a retryable
+    minibatch error (avoids multiplying retry loops)."""
+
+
+def is_collective_error(e: BaseException) -> bool:
+    msg = str(e).lower()
+    return any(k in msg for k in (
+        "nccl", "rccl", "connection", "timeout", "timed out",
+        "process group", "store", "socket",
+    ))
+
+
 class CommunicatorManager:
     def __init__(
         self,
@@ -74,17 +88,39 @@ class CommunicatorManager:
     def ensure_communicator(self, poll_interval: float = 1.0) -> bool:
         """Poll the master; (re)build the process group if the rendezvous
         generation moved. Returns True if the communicator was re-formed
-        (caller must re-broadcast state from rank 0)."""
+        (caller must re-broadcast state from rank 0).
+
+        A generation on which THIS worker saw a collective/bootstrap
+        failure is never rebuilt (its store keys may be half-written) —
+        handle_collective_failure() asks the master to bump the
+        generation, and this loop waits for the new one."""
         assert self._master_client is not None, "elastic mode needs a master"
         while True:
             info = self._master_client.get_comm_rank(self._worker_host)
-            if info["rank_id"] >= 0 and info["world_size"] > 0:
+            if (
+                info["rank_id"] >= 0
+                and info["world_size"] > 0
+                and info["rendezvous_id"] != self._failed_gen
+            ):
                 break
             time.sleep(poll_interval)
         if info["rendezvous_id"] == self.rendezvous_id and self._pg_alive():
             return False
         self._rebuild(info)
         return True
+
+    def handle_collective_failure(self) -> None:
+        """Collective or bootstrap failed: destroy local state, poison the
+        current generation, and ask the master for a fresh one."""
+        failed = max(self.rendezvous_id, self._attempted_gen)
+        self.teardown()
+        self._failed_gen = failed
+        self.rendezvous_id = -1
+        if self._master_client is not None:
+            try:
+                self._master_client.report_training_loop_status("reset")
+            except Exception:  # noqa: BLE001 - master may be restarting
+                logger.warning("reset request to master failed")
 
     def _pg_alive(self) -> bool:
         return dist.is_initialized()
@@ -94,6 +130,7 @@ class CommunicatorManager:
         rank = info["rank_id"]
         world = info["world_size"]
         rdzv = info["rendezvous_id"]
+        self._attempted_gen = rdzv
         host, port = self._master_client.rendezvous_addr(info)
         logger.info(
             "Building communicator gen=%d rank=%d world=%d via %s:%d",
@@ -118,6 +155,9 @@ class CommunicatorManager:
         self.rank = rank
         self.world_size = world
         self.need_broadcast = True
+
+    _failed_gen = -1
+    _attempted_gen = -1
 
     def teardown(self) -> None:
         """Abort and destroy the current process group (safe to call when

@@ -17,7 +17,11 @@ from typing import Callable, Optional
 
 import torch
 
-from elasticdl_amd.collective.communicator import CommunicatorManager
+from elasticdl_amd.collective.communicator import (
+    CollectiveFailureError,
+    CommunicatorManager,
+    is_collective_error,
+)
 from elasticdl_amd.collective.distributed_optimizer import DistributedOptimizer
 from elasticdl_amd.common.constants import MAX_ALLREDUCE_RETRY_NUM
 from elasticdl_amd.common.log_utils import default_logger as logger
@@ -101,17 +105,15 @@ class ElasticAllReduceController:
                     self.data_shard_service.report_batch_done()
                     return out
                 except RuntimeError as e:
-                    msg = str(e).lower()
-                    if any(k in msg for k in ("nccl", "rccl", "connection",
-                                              "timeout", "process group")):
+                    if is_collective_error(e):
                         logger.warning(
                             "Collective failed (%s); re-init (%d/%d)",
                             e, attempt + 1, MAX_ALLREDUCE_RETRY_NUM,
                         )
-                        self.comm.teardown()
-                        time.sleep(3)
+                        self.comm.handle_collective_failure()
+                        time.sleep(1)
                         continue
                     raise
-            raise RuntimeError("elastic_run: retries exhausted")
+            raise CollectiveFailureError("elastic_run: retries exhausted")
 
         return wrapped

@@ -112,6 +112,17 @@ class ElasticRendezvousServer:
                     self._next_hosts,
                 )
 
+    def force_reset(self) -> None:
+        """Stage a generation bump without membership change — used when a
+        worker reports a collective/bootstrap failure so the whole world
+        re-forms under a FRESH key prefix (rebuilding the same generation
+        would reread half-written bootstrap keys)."""
+        with self._lock:
+            if self._next_hosts is None:
+                self._next_hosts = copy.deepcopy(self._cur_hosts)
+            logger.info("Rendezvous: reset requested (next world %s)",
+                        self._next_hosts)
+
     # ----------------------------------------------------------------- query
     def get_comm_rank(self, worker_host: str) -> dict:
         """One-stop poll for workers: rank/world/rendezvous_id/store addr."""

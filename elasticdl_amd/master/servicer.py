@@ -28,6 +28,7 @@ class TrainingLoopStatus:
     START = "start"
     END = "end"
     PENDING = "pending"
+    RESET = "reset"  # collective failed: re-form under a new generation
 
 
 class MasterServicer:
@@ -130,6 +131,8 @@ class MasterServicer:
             self._rendezvous_server.add_worker(host)
         elif status == TrainingLoopStatus.END:
             self._rendezvous_server.remove_worker(host)
+        elif status == TrainingLoopStatus.RESET:
+            self._rendezvous_server.force_reset()
         return {}
 
     def report_version(self, req: dict) -> dict:

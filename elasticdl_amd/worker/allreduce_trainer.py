@@ -18,7 +18,11 @@ from typing import Optional
 
 import torch
 
-from elasticdl_amd.collective.communicator import CommunicatorManager
+from elasticdl_amd.collective.communicator import (
+    CollectiveFailureError,
+    CommunicatorManager,
+    is_collective_error,
+)
 from elasticdl_amd.collective.distributed_optimizer import DistributedOptimizer
 from elasticdl_amd.common.constants import MAX_ALLREDUCE_RETRY_NUM
 from elasticdl_amd.common.log_utils import default_logger as logger
@@ -115,18 +119,16 @@ class AllReduceTrainer(Trainer):
                 self.init_communicator_if_needed(force=attempt > 0)
                 return self._train_once(batch)
             except RuntimeError as e:
-                msg = str(e).lower()
-                if any(k in msg for k in ("nccl", "rccl", "connection",
-                                          "timeout", "process group")):
+                if is_collective_error(e):
                     logger.warning(
                         "Collective failed (%s); re-initializing (%d/%d)",
                         e, attempt + 1, MAX_ALLREDUCE_RETRY_NUM,
                     )
-                    self.comm.teardown()
-                    time.sleep(3)
+                    self.comm.handle_collective_failure()
+                    time.sleep(1)
                     continue
                 raise
-        raise RuntimeError("allreduce retries exhausted")
+        raise CollectiveFailureError("allreduce retries exhausted")
 
     def _train_once(self, batch):
         x, y = self._feed(batch)

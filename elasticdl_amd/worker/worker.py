@@ -119,6 +119,15 @@ class Worker:
                     return loss
                 return self.trainer.evaluate_minibatch(batch)
             except Exception as e:  # noqa: BLE001 - retried
+                from elasticdl_amd.collective.communicator import (
+                    CollectiveFailureError,
+                )
+
+                if isinstance(e, CollectiveFailureError):
+                    # the trainer already exhausted its re-init budget;
+                    # escalate to task failure instead of multiplying
+                    # retries 64x
+                    raise
                 err = e
                 if attempt < 2:
                     logger.warning("minibatch failed (%s); retrying", e)

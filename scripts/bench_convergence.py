@@ -1,0 +1,138 @@
+#!/usr/bin/env python3
+"""Convergence parity under elasticity (reference claim:
+docs/benchmark/report_cn.md:108-120 — elastic 4<->8 worker curves are
+indistinguishable from fixed-size gang runs).
+
+Runs the same seeded workload three ways on CPU (gloo):
+  A. fixed 2 workers
+  B. fixed 1 worker
+  C. elastic: start 2 workers, kill one mid-run (master requeues its
+     shards; survivor re-forms the world and finishes)
+and reports final losses. Synthetic separable data so loss is a clean
+signal.
+
+    python scripts/bench_convergence.py
+"""
+
+import json
+import os
+import signal
+import subprocess
+import sys
+import tempfile
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+ZOO = os.path.join(tempfile.gettempdir(), "edl_conv_zoo.py")
+
+ZOO_SRC = '''
+import torch
+import torch.nn as nn
+
+
+def custom_model():
+    torch.manual_seed(7)
+    return nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 1),
+                         nn.Flatten(0))
+
+
+def loss(outputs, labels):
+    return nn.functional.binary_cross_entropy_with_logits(
+        outputs.float(), labels.float())
+
+
+def optimizer(model=None):
+    return ("sgd", "learning_rate=0.05;momentum=0.9")
+
+
+def eval_metrics_fn():
+    return {"accuracy": lambda o, l: ((o > 0).long() == l.long()).float().mean()}
+
+
+def feed(batch, device, dtype=None):
+    x, y = batch
+    return x.to(device), y.to(device)
+
+
+def custom_data_reader(data_origin=""):
+    from elasticdl_amd.data.reader import SyntheticReader
+
+    size = int(data_origin.split(":")[1]) if ":" in data_origin else 2048
+    w = torch.randn(16, generator=torch.Generator().manual_seed(3))
+
+    def sample(i):
+        g = torch.Generator().manual_seed(i)
+        x = torch.randn(16, generator=g)
+        y = (x @ w > 0).long()
+        return x, y
+
+    return SyntheticReader(size, sample, name="sep-synthetic")
+'''
+
+
+def run(num_workers, kill_one=False, records=4096):
+    with open(ZOO, "w") as f:
+        f.write(ZOO_SRC)
+    tmp = tempfile.mkdtemp(prefix="edl-conv-")
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", ZOO,
+        "--distribution_strategy", "AllreduceStrategy",
+        "--num_workers", str(num_workers),
+        "--minibatch_size", "32",
+        "--num_minibatches_per_task", "4",
+        "--num_epochs", "2",
+        "--training_data", f"synthetic:{records}",
+        "--device", "cpu",
+        "--checkpoint_dir", tmp,
+        "--log_loss_steps", "5",
+        "--pod_manager", "local",
+    ]
+    env = dict(os.environ, PYTHONPATH=REPO, EDL_PG_TIMEOUT_SEC="20")
+    p = subprocess.Popen(cmd, env=env, cwd=REPO, stdout=subprocess.PIPE,
+                         stderr=subprocess.STDOUT, text=True)
+    if kill_one:
+        time.sleep(25)
+        pidfile = os.path.join(tmp, "logs", "worker-0.pid")
+        if os.path.exists(pidfile):
+            try:
+                os.kill(int(open(pidfile).read()), signal.SIGKILL)
+                print("[convergence] killed worker-0", flush=True)
+            except ProcessLookupError:
+                pass
+    out, _ = p.communicate(timeout=900)
+    losses = []
+    for line in out.splitlines():
+        if " loss " in line:
+            try:
+                losses.append(float(line.split(" loss ")[1].split(" ")[0]))
+            except (ValueError, IndexError):
+                pass
+    final = sum(losses[-5:]) / max(len(losses[-5:]), 1) if losses else None
+    return p.returncode, final, len(losses)
+
+
+def main():
+    results = {}
+    for name, kw in [
+        ("fixed_2_workers", dict(num_workers=2)),
+        ("fixed_1_worker", dict(num_workers=1)),
+        ("elastic_2_minus_1", dict(num_workers=2, kill_one=True)),
+    ]:
+        rc, final, n = run(**kw)
+        results[name] = {"exit": rc, "final_loss": final, "logged_steps": n}
+        print(f"[convergence] {name}: {results[name]}", flush=True)
+    print(json.dumps({"metric": "convergence_parity", **results}), flush=True)
+    ok = all(v["exit"] == 0 and v["final_loss"] is not None
+             for v in results.values())
+    losses = [v["final_loss"] for v in results.values()]
+    spread = max(losses) - min(losses)
+    print(f"[convergence] final-loss spread across runs: {spread:.4f}",
+          flush=True)
+    return 0 if ok else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
