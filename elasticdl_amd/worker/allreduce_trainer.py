@@ -177,6 +177,20 @@ class AllReduceTrainer(Trainer):
         self.comm.teardown()
 
     def on_training_start(self) -> None:
+        import os
+
         from elasticdl_amd.master.servicer import TrainingLoopStatus
 
         self.mc.report_training_loop_status(TrainingLoopStatus.START)
+        # Optional co-start barrier (benchmarks/experiments): wait until at
+        # least EDL_MIN_WORLD workers are in the communicator before the
+        # first batch. Default elastic behavior (train with whoever is
+        # present) is unchanged when unset.
+        min_world = int(os.environ.get("EDL_MIN_WORLD", "0"))
+        if min_world > 1:
+            deadline = time.monotonic() + 120
+            while time.monotonic() < deadline:
+                self.init_communicator_if_needed(force=True)
+                if self.comm.world_size >= min_world:
+                    break
+                time.sleep(1)

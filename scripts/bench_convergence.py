@@ -34,8 +34,9 @@ import torch.nn as nn
 
 def custom_model():
     torch.manual_seed(7)
-    return nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 1),
-                         nn.Flatten(0))
+    return nn.Sequential(nn.Linear(16, 512), nn.ReLU(),
+                         nn.Linear(512, 512), nn.ReLU(),
+                         nn.Linear(512, 1), nn.Flatten(0))
 
 
 def loss(outputs, labels):
@@ -72,7 +73,7 @@ def custom_data_reader(data_origin=""):
 '''
 
 
-def run(num_workers, kill_one=False, records=4096):
+def run(num_workers, kill_one=False, records=8192):
     with open(ZOO, "w") as f:
         f.write(ZOO_SRC)
     tmp = tempfile.mkdtemp(prefix="edl-conv-")
@@ -90,11 +91,12 @@ def run(num_workers, kill_one=False, records=4096):
         "--log_loss_steps", "5",
         "--pod_manager", "local",
     ]
-    env = dict(os.environ, PYTHONPATH=REPO, EDL_PG_TIMEOUT_SEC="20")
+    env = dict(os.environ, PYTHONPATH=REPO, EDL_PG_TIMEOUT_SEC="20",
+               EDL_MIN_WORLD=str(num_workers))
     p = subprocess.Popen(cmd, env=env, cwd=REPO, stdout=subprocess.PIPE,
                          stderr=subprocess.STDOUT, text=True)
     if kill_one:
-        time.sleep(25)
+        time.sleep(20)
         pidfile = os.path.join(tmp, "logs", "worker-0.pid")
         if os.path.exists(pidfile):
             try:
@@ -103,14 +105,23 @@ def run(num_workers, kill_one=False, records=4096):
             except ProcessLookupError:
                 pass
     out, _ = p.communicate(timeout=900)
+    # loss lines are in the per-worker log files (master redirects)
     losses = []
-    for line in out.splitlines():
-        if " loss " in line:
-            try:
-                losses.append(float(line.split(" loss ")[1].split(" ")[0]))
-            except (ValueError, IndexError):
-                pass
-    final = sum(losses[-5:]) / max(len(losses[-5:]), 1) if losses else None
+    logdir = os.path.join(tmp, "logs")
+    if os.path.isdir(logdir):
+        for fn in sorted(os.listdir(logdir)):
+            if fn.startswith("worker") and fn.endswith(".log"):
+                for line in open(os.path.join(logdir, fn), errors="replace"):
+                    if " loss " in line:
+                        try:
+                            step = int(line.split("step ")[1].split(" ")[0])
+                            loss = float(line.split(" loss ")[1].split(" ")[0])
+                            losses.append((step, loss))
+                        except (ValueError, IndexError):
+                            pass
+    losses.sort()
+    tail = [v for _, v in losses[-8:]]
+    final = sum(tail) / len(tail) if tail else None
     return p.returncode, final, len(losses)
 
 
