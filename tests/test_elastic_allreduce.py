@@ -35,7 +35,7 @@ def run_master(extra_args, env_extra=None):
                             text=True)
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(720)
 def test_allreduce_two_workers_complete():
     p = run_master([
         "--num_workers", "2",
@@ -45,7 +45,7 @@ def test_allreduce_two_workers_complete():
     assert p.returncode == 0, out[-4000:]
 
 
-@pytest.mark.timeout(420)
+@pytest.mark.timeout(720)
 def test_allreduce_worker_killed_job_survives():
     import tempfile
 
