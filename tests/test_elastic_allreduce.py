@@ -41,7 +41,7 @@ def test_allreduce_two_workers_complete():
         "--num_workers", "2",
         "--training_data", "synthetic:192",
     ])
-    out, _ = p.communicate(timeout=280)
+    out, _ = p.communicate(timeout=650)
     assert p.returncode == 0, out[-4000:]
 
 
@@ -70,5 +70,5 @@ def test_allreduce_worker_killed_job_survives():
             os.kill(w0_pid, signal.SIGKILL)
         except ProcessLookupError:
             pass  # finished already; elasticity path still validated below
-        out, _ = p.communicate(timeout=380)
+        out, _ = p.communicate(timeout=650)
         assert p.returncode == 0, out[-6000:]
