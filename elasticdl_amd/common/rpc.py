@@ -121,12 +121,14 @@ class RpcClient:
         data = codec.encode(message if message is not None else {})
         return codec.decode(self._stub(service, method)(data, timeout=timeout))
 
-    def call_future(self, service: str, method: str, message: Any = None):
+    def call_future(self, service: str, method: str, message: Any = None,
+                    timeout: Optional[float] = 300.0):
         """Async fan-out variant (reference: ps_client futures); returns a
         future whose .result() must be passed through codec.decode by
-        ``resolve``."""
+        ``resolve``. A default deadline keeps workers from hanging forever
+        on a dead PS (the master's job-failure path then kicks in)."""
         data = codec.encode(message if message is not None else {})
-        return self._stub(service, method).future(data)
+        return self._stub(service, method).future(data, timeout=timeout)
 
     @staticmethod
     def resolve(future) -> Any:

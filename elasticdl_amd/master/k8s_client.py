@@ -54,6 +54,25 @@ def parse_volume(spec: str) -> List[Dict[str, str]]:
     return out
 
 
+class ClusterSpec:
+    """Bespoke-cluster pod mutation hook (reference:
+    elasticdl_client/common/k8s_client.py:106-219): a python module/file
+    exposing ``patch_pod(pod, pod_type) -> pod`` applied to every pod spec
+    before creation."""
+
+    def __init__(self, spec: str = ""):
+        self._mod = None
+        if spec:
+            from elasticdl_amd.utils.model_utils import load_module
+
+            self._mod = load_module(spec)
+
+    def patch_pod(self, pod, pod_type: str):
+        if self._mod is not None and hasattr(self._mod, "patch_pod"):
+            return self._mod.patch_pod(pod, pod_type) or pod
+        return pod
+
+
 class Client:
     def __init__(
         self,

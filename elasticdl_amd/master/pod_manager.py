@@ -52,6 +52,9 @@ class PodManager:
             job_name=args.job_name,
             image_name=args.image_name,
         )
+        from elasticdl_amd.master.k8s_client import ClusterSpec
+
+        self.cluster_spec = ClusterSpec(getattr(args, "cluster_spec", ""))
         self._lock = threading.Lock()
         self.pods: Dict[str, PodInfo] = {}
         self._next_worker_id = 0
@@ -128,6 +131,7 @@ class PodManager:
             restart_policy=self.args.restart_policy,
             owner_pod=owner,
         )
+        pod = self.cluster_spec.patch_pod(pod, pod_type)
         with self._lock:
             self.pods[name] = PodInfo(name, pod_type, index)
         if not self.k8s.create_pod(pod):
