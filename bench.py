@@ -147,7 +147,7 @@ def bench_ps_model(args, rank, world, local_rank):
     from elasticdl_amd.collective.distributed_optimizer import DistributedOptimizer
     from elasticdl_amd.common.tensor_utils import merge_indexed_slices
     from elasticdl_amd.layers.embedding import find_edl_embeddings
-    from elasticdl_amd.models import deepfm, wide_deep
+    from elasticdl_amd.models import dcn, deepfm, wide_deep
     from elasticdl_amd.ps.engine import PSEngine
     from elasticdl_amd.ps.sharded import ShardedPSEngine
 
@@ -157,13 +157,14 @@ def bench_ps_model(args, rank, world, local_rank):
         torch.cuda.set_device(device)
     torch.manual_seed(1234)  # same dense init on all ranks
 
-    zoo = deepfm if args.model == "deepfm" else wide_deep
+    zoo = {"deepfm": deepfm, "wide_deep": wide_deep, "dcn": dcn}[args.model]
     if args.model == "deepfm":
         model = deepfm.DeepFM(max_rows=args.table_rows)
-        batch_fn = lambda s: deepfm.synthetic_batch(args.batch_size, seed=s)
+    elif args.model == "dcn":
+        model = dcn.DCN(max_rows=args.table_rows)
     else:
         model = wide_deep.WideDeep(max_rows=args.table_rows)
-        batch_fn = lambda s: wide_deep.synthetic_batch(args.batch_size, seed=s)
+    batch_fn = lambda s: zoo.synthetic_batch(args.batch_size, seed=s)
     model = model.to(device)
     if device.type == "cuda":
         model = model.to(torch.bfloat16)
@@ -245,7 +246,7 @@ def main():
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--model", default="resnet50",
-                    choices=["resnet50", "deepfm", "wide_deep"])
+                    choices=["resnet50", "deepfm", "wide_deep", "dcn"])
     ap.add_argument("--batch-size", "--batch_size", type=int, default=None)
     ap.add_argument("--image-size", "--image_size", type=int, default=224)
     ap.add_argument("--num-classes", "--num_classes", type=int, default=1000)
