@@ -189,6 +189,11 @@ class DistributedOptimizer:
         for b in self.buckets:
             b.grad_flat.zero_()
             b.assign_grad_views()  # backward may have replaced .grad
+            # drop in-flight allreduce handles — after an elastic re-init
+            # they reference a destroyed process group and waiting on them
+            # in step() would raise/hang (the retried backward re-issues)
+            b.work = None
+            b.ready_count = 0
         self._pass_count = 0
 
     # --------------------------------------------------------------- step

@@ -218,6 +218,9 @@ class PSEngine:
             self.optimizer.apply_sparse(table, s.values, s.ids, lr_mult)
 
     def _notify(self, version: int) -> None:
+        if version % 256 == 0:  # periodic overflow check (one .item() sync)
+            for t in self.tables.values():
+                t.check_health()
         for fn in self.version_listeners:
             fn(version)
 
