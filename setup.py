@@ -8,7 +8,7 @@ boxes (it is git-ignored but not gpurun-ignored).
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -33,7 +33,10 @@ ext = CUDAExtension(
 setup(
     name="elasticdl_amd",
     version="0.1.0",
-    packages=["elasticdl_amd"],
+    packages=find_packages(include=["elasticdl_amd", "elasticdl_amd.*"]),
+    entry_points={
+        "console_scripts": ["elasticdl=elasticdl_amd.client.main:main"],
+    },
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension},
 )
