@@ -89,6 +89,11 @@ class ShardedPSEngine:
     """One PSEngine shard per rank + all-to-all row exchange."""
 
     def __init__(self, local_engine: PSEngine, group=None):
+        # in-job sharding is async-only, like the reference's Go PS
+        # (go/pkg/ps/server.go:177): sync accumulation would need a
+        # cross-rank grads_to_wait barrier that the all-to-all data plane
+        # deliberately avoids
+        assert local_engine.use_async, "ShardedPSEngine requires use_async"
         self.local = local_engine
         self.group = group
         self.world = (
