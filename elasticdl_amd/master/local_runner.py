@@ -58,9 +58,33 @@ class LocalProcessManager:
         self._callbacks.append(cb)
 
     # ------------------------------------------------------------ lifecycle
+    @staticmethod
+    def _gpu_assignment(pod_id: int) -> Optional[str]:
+        """Round-robin local pods over the node's GPUs (one process per
+        GPU, the MI355X scaling model). Honors a pre-set visibility list."""
+        try:
+            import torch
+
+            if not torch.cuda.is_available():
+                return None
+            visible = os.environ.get("HIP_VISIBLE_DEVICES")
+            if visible:
+                devices = [d for d in visible.split(",") if d != ""]
+            else:
+                devices = [str(i) for i in range(torch.cuda.device_count())]
+            if len(devices) <= 1:
+                return None
+            return devices[pod_id % len(devices)]
+        except Exception:  # noqa: BLE001 - placement is best-effort
+            return None
+
     def _spawn(self, name: str, cmd: List[str], pod_type: str, pod_id: int,
                extra_env: Dict[str, str]) -> LocalProcess:
         env = dict(os.environ)
+        if pod_type in ("worker", "ps"):
+            gpu = self._gpu_assignment(pod_id)
+            if gpu is not None:
+                env["HIP_VISIBLE_DEVICES"] = gpu
         repo_root = os.path.dirname(
             os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
         )
