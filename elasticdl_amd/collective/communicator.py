@@ -48,7 +48,9 @@ class CommunicatorManager:
     ):
         self._master_client = master_client
         self._worker_host = worker_host
-        self._backend = backend or (
+        # EDL_BACKEND overrides (e.g. gloo to run N>1 workers against a
+        # single shared GPU, where RCCL refuses duplicate devices)
+        self._backend = backend or os.environ.get("EDL_BACKEND") or (
             "nccl" if torch.cuda.is_available() else "gloo"
         )
         if init_timeout is None:

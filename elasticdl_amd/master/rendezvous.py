@@ -80,10 +80,11 @@ class ElasticRendezvousServer:
                 return
             if self._next_hosts is None:
                 self._next_hosts = copy.deepcopy(self._cur_hosts)
-            # after training started, never resurrect an empty world
-            # (reference: rendezvous_server.py add_worker)
-            if self._rendezvous_id > 0 and not self._next_hosts:
-                return
+            # NOTE: the reference refuses to resurrect an empty world
+            # (rendezvous_server.py add_worker) because its master stops
+            # the job when every worker dies; here relaunched workers are
+            # first-class and MUST be able to re-form a world after a
+            # total wipeout, so the refusal is intentionally dropped.
             if worker_host not in self._next_hosts:
                 self._next_hosts.append(worker_host)
                 logger.info(
@@ -127,18 +128,16 @@ class ElasticRendezvousServer:
     def get_comm_rank(self, worker_host: str) -> dict:
         """One-stop poll for workers: rank/world/rendezvous_id/store addr."""
         with self._lock:
+            self._mark_ready(worker_host)
             if self._next_hosts is not None and self._cur_completed:
                 time.sleep(self._flip_delay_sec)
                 self._flip()
-            if worker_host not in self._cur_hosts:
-                rank = -1
-            else:
-                if not self._cur_completed:
-                    self._ready_hosts.add(worker_host)
-                    if self._ready_hosts >= set(self._cur_hosts) - self._dead_hosts:
-                        self._cur_completed = True
-                        self._ready_hosts = set()
-                rank = self._cur_hosts.index(worker_host)
+                self._mark_ready(worker_host)
+            rank = (
+                self._cur_hosts.index(worker_host)
+                if worker_host in self._cur_hosts
+                else -1
+            )
             return {
                 "rank_id": rank,
                 "world_size": len(self._cur_hosts),
@@ -146,11 +145,26 @@ class ElasticRendezvousServer:
                 "rendezvous_port": self._port or 0,
             }
 
+    def _mark_ready(self, worker_host: str) -> None:
+        """Update current-world readiness. Evaluated on EVERY poll (not
+        just members'): when all remaining members of the current world
+        are dead, any poller — e.g. a relaunched worker waiting to join —
+        completes it so the staged world can flip."""
+        if self._cur_completed:
+            return
+        if worker_host in self._cur_hosts:
+            self._ready_hosts.add(worker_host)
+        if self._ready_hosts >= set(self._cur_hosts) - self._dead_hosts:
+            self._cur_completed = True
+            self._ready_hosts = set()
+
     def _flip(self) -> None:
         self._cur_hosts = self._next_hosts
         self._next_hosts = None
         self._rendezvous_id += 1
-        self._cur_completed = False
+        # an empty world can never report readiness — it is trivially
+        # complete, so the next staged world can flip immediately
+        self._cur_completed = not self._cur_hosts
         self._ready_hosts = set()
         self._dead_hosts = set()
         logger.info(

@@ -58,13 +58,55 @@ def test_removal_before_ready_coalesces():
     r.add_worker("w1")
     r.add_worker("w2")
     assert r.get_comm_rank("w0")["world_size"] == 3
-    # w1 dies before the world completes; w2 removal staged
+    # w1 dies before the world completes: the dead member is exempt from
+    # readiness, so once the remaining members have polled the staged
+    # world [w0, w2] flips (gen 2)
     r.remove_worker("w1")
-    # remaining workers keep polling; current world can't complete (w1 gone)
-    # -> flip happens only once cur world is considered done. Force by
-    # having all *remaining* hosts report; then the staged world flips when
-    # current completes.
-    assert r.get_comm_rank("w2")["rendezvous_id"] == 1
+    info = r.get_comm_rank("w2")
+    assert info["rendezvous_id"] == 2
+    assert info["world_size"] == 2
+    assert r.get_comm_rank("w0")["rank_id"] == 0
+
+
+def test_world_resurrects_after_total_wipeout():
+    """All workers die, a relaunched worker must be able to re-form the
+    world (deviation from the reference, which refuses — its master would
+    have killed the job; ours relaunches workers)."""
+    r = make_rdzv()
+    r.add_worker("w0")
+    r.add_worker("w1")
+    r.get_comm_rank("w0")
+    r.get_comm_rank("w1")  # world 1 complete
+    r.remove_worker("w0")
+    r.remove_worker("w1")
+    # any poll flips to the empty world
+    assert r.get_comm_rank("w0")["rank_id"] == -1
+    assert r.world_size() == 0
+    # relaunched worker joins; empty world is trivially complete -> flip
+    r.add_worker("w2")
+    info = r.get_comm_rank("w2")
+    assert info["rank_id"] == 0
+    assert info["world_size"] == 1
+    assert info["rendezvous_id"] >= 3
+
+
+def test_relaunched_worker_joins_when_all_members_dead():
+    """The exact flake scenario: every member of the current world died
+    (removed) before reporting ready; a relaunched NON-member's poll must
+    complete the world so the staged one flips."""
+    r = make_rdzv()
+    r.add_worker("w0")
+    r.add_worker("w1")
+    assert r.get_comm_rank("w0")["world_size"] == 2  # gen 1 formed
+    # neither member ever polls again (both die); master stages removals
+    r.remove_worker("w1")
+    r.remove_worker("w0")
+    r.add_worker("w2")  # relaunched worker
+    info = r.get_comm_rank("w2")  # its poll alone must drive the flip(s)
+    if info["rank_id"] < 0:  # at most one more poll needed
+        info = r.get_comm_rank("w2")
+    assert info["rank_id"] == 0
+    assert info["world_size"] == 1
 
 
 def test_real_tcpstore_start():
