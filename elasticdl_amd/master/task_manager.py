@@ -76,6 +76,12 @@ class TaskManager:
         self.total_records = sum(s[2] - s[1] for s in self._training_shards)
         self.failed_records = 0
 
+        # worker-driven jobs (SDK path): no shards at startup — the job is
+        # not "finished" until a worker reports training params or tasks
+        # are injected (reference: master waits on workers)
+        self._awaiting_tasks = not (
+            training_shards or evaluation_shards or prediction_shards
+        )
         self._train_end_callback_emitted = False
         self._train_end_callback_done = False
         self._eval_todo_count = 0
@@ -153,6 +159,7 @@ class TaskManager:
             self._shuffle = shuffle
             self._shuffle_shards = shuffle_shards
             self.total_records = dataset_size
+            self._awaiting_tasks = False
             self._create_training_tasks()
             logger.info(
                 "TaskManager: %d worker-defined training tasks (%d records)",
@@ -169,12 +176,16 @@ class TaskManager:
             # evaluation tasks go to the head so they interleave promptly
             self._todo = tasks + self._todo
             self._eval_todo_count += len(tasks)
+            if tasks:
+                self._awaiting_tasks = False
             return len(tasks)
 
     def create_prediction_tasks(self) -> int:
         with self._lock:
             tasks = self._shards_to_tasks(self._prediction_shards, TaskType.PREDICTION)
             self._todo.extend(tasks)
+            if tasks:
+                self._awaiting_tasks = False
             return len(tasks)
 
     def create_train_end_callback_task(self) -> None:
@@ -345,6 +356,8 @@ class TaskManager:
         with self._lock:
             if self._max_step and self._completed_steps >= self._max_step:
                 return True
+            if self._awaiting_tasks:
+                return False
             if self._todo or self._doing:
                 return False
             if self._training_shards and self._epoch + 1 < self._num_epochs:
