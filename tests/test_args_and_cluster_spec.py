@@ -1,0 +1,47 @@
+import os
+
+from elasticdl_amd.common.args import (
+    build_arguments_from_parsed_result,
+    parse_master_args,
+    parse_model_params,
+)
+
+
+def test_args_round_trip():
+    argv = [
+        "--model_def", "deepfm", "--num_workers", "3",
+        "--distribution_strategy", "AllreduceStrategy",
+        "--minibatch_size", "64", "--use_async", "False",
+        "--shuffle", "true",
+    ]
+    args = parse_master_args(argv)
+    assert args.num_workers == 3
+    assert args.use_async is False
+    assert args.shuffle is True
+    rebuilt = build_arguments_from_parsed_result(args)
+    args2 = parse_master_args(rebuilt)
+    assert vars(args2) == vars(args)
+
+
+def test_parse_model_params():
+    p = parse_model_params("num_fields=39;hidden=[64, 32];name=abc")
+    assert p == {"num_fields": 39, "hidden": [64, 32], "name": "abc"}
+
+
+def test_cluster_spec_hook(tmp_path):
+    mod = tmp_path / "spec.py"
+    mod.write_text(
+        "def patch_pod(pod, pod_type):\n"
+        "    pod.patched = pod_type\n"
+        "    return pod\n"
+    )
+    from types import SimpleNamespace as NS
+
+    from elasticdl_amd.master.k8s_client import ClusterSpec
+
+    cs = ClusterSpec(str(mod))
+    pod = NS()
+    out = cs.patch_pod(pod, "worker")
+    assert out.patched == "worker"
+    # empty spec is a no-op
+    assert ClusterSpec("").patch_pod(pod, "ps") is pod
