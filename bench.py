@@ -61,7 +61,13 @@ def bench_resnet50(args, rank, world, local_rank):
     device = torch.device("cuda", local_rank) if torch.cuda.is_available() else torch.device("cpu")
     if device.type == "cuda":
         torch.cuda.set_device(device)
-        # MIOpen conv autotune (skippable for clean profiling runs)
+        # MIOpen conv autotune (skippable for clean profiling runs).
+        # Per-rank user DBs: N ranks autotuning concurrently must not
+        # serialize on the shared DB's file lock.
+        if world > 1:
+            os.environ.setdefault(
+                "MIOPEN_USER_DB_PATH", f"/tmp/miopen-rank{local_rank}"
+            )
         torch.backends.cudnn.benchmark = os.environ.get("EDL_NO_AUTOTUNE") != "1"
 
     torch.manual_seed(1234)
