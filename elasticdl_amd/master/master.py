@@ -268,6 +268,14 @@ class Master:
         self.task_manager.stop()
         if self.pod_manager is not None:
             self.pod_manager.stop()
+            # advertise final job status on the master pod for external
+            # observers (reference: pod_manager.py:444-448)
+            k8s = getattr(self.pod_manager, "k8s", None)
+            if k8s is not None:
+                k8s.patch_labels_to_pod(
+                    k8s.get_master_pod_name(),
+                    {"status": "Finished" if self.exit_code == 0 else "Failed"},
+                )
         if self.server is not None:
             self.server.stop(1)
 
