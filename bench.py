@@ -68,7 +68,11 @@ def bench_resnet50(args, rank, world, local_rank):
             os.environ.setdefault(
                 "MIOPEN_USER_DB_PATH", f"/tmp/miopen-rank{local_rank}"
             )
-        torch.backends.cudnn.benchmark = os.environ.get("EDL_NO_AUTOTUNE") != "1"
+        # Exhaustive conv search costs minutes of warmup for ~5% steady
+        # state; only worth it when the caller budgets enough warmup.
+        torch.backends.cudnn.benchmark = (
+            os.environ.get("EDL_NO_AUTOTUNE") != "1" and args.warmup >= 8
+        )
 
     torch.manual_seed(1234)
     model = resnet.resnet50(num_classes=args.num_classes)
