@@ -122,6 +122,13 @@ class DistributedOptimizer:
         self._native = device.type == "cuda"
         if grad_dtype is None:
             grad_dtype = params[0].dtype
+        mixed = {p.dtype for p in params}
+        if len(mixed) > 1:
+            raise ValueError(
+                "DistributedOptimizer buckets require a uniform parameter "
+                f"dtype, got {sorted(str(d) for d in mixed)} — cast the "
+                "model (e.g. model.to(torch.bfloat16)) first"
+            )
         self.grad_dtype = grad_dtype
 
         cap = int(bucket_cap_mb * 1024 * 1024 / grad_dtype.itemsize)
