@@ -211,6 +211,10 @@ class TaskManager:
         with self._lock:
             if self._max_step and self._completed_steps >= self._max_step:
                 return Task(task_id=0, shard=None, type=TaskType.NONE)
+            if self._awaiting_tasks:
+                # worker-driven job: tasks arrive once a worker reports
+                # training params — poll again
+                return Task(task_id=0, shard=None, type=TaskType.WAIT)
             if not self._todo:
                 if self._maybe_start_next_epoch():
                     pass  # fall through with refilled todo

@@ -1,0 +1,119 @@
+"""Small-unit coverage: timing, rpc error paths, codec fuzz, client
+degradation, worker WAIT continuation."""
+
+import random
+import threading
+import time
+
+import grpc
+import pytest
+import torch
+
+from elasticdl_amd.common import codec, rpc
+from elasticdl_amd.common.task import TaskType
+from elasticdl_amd.common.timing import Timing
+
+
+def test_timing_sections():
+    t = Timing(enabled=True)
+    t.start_record_time("a")
+    time.sleep(0.01)
+    t.end_record_time("a")
+    report = t.report_timing(reset=True)
+    assert "a:" in report
+    assert t.acc == {}
+    # disabled timing is a no-op
+    t2 = Timing(enabled=False)
+    t2.start_record_time("x")
+    t2.end_record_time("x")
+    assert t2.acc == {}
+
+
+def test_codec_fuzz_roundtrip():
+    rng = random.Random(0)
+
+    def gen(depth=0):
+        kind = rng.randrange(6 if depth < 3 else 4)
+        if kind == 0:
+            return rng.randint(-10**12, 10**12)
+        if kind == 1:
+            return rng.random()
+        if kind == 2:
+            return "".join(chr(rng.randrange(32, 1000)) for _ in range(8))
+        if kind == 3:
+            return torch.randn(rng.randrange(0, 5), rng.randrange(1, 4))
+        if kind == 4:
+            return [gen(depth + 1) for _ in range(rng.randrange(4))]
+        return {f"k{i}": gen(depth + 1) for i in range(rng.randrange(4))}
+
+    for _ in range(25):
+        msg = {"payload": gen()}
+        out = codec.decode(codec.encode(msg))
+
+        def eq(a, b):
+            if isinstance(a, torch.Tensor):
+                return torch.equal(a, b)
+            if isinstance(a, (list, tuple)):
+                return len(a) == len(b) and all(eq(x, y) for x, y in zip(a, b))
+            if isinstance(a, dict):
+                return a.keys() == b.keys() and all(eq(a[k], b[k]) for k in a)
+            if isinstance(a, float):
+                return abs(a - b) < 1e-12
+            return a == b
+
+        assert eq(msg, out)
+
+
+def test_rpc_handler_error_propagates_as_internal():
+    def boom(req):
+        raise ValueError("nope")
+
+    server = rpc.start_server("127.0.0.1:0", {"S": {"boom": boom}})
+    try:
+        client = rpc.RpcClient(f"127.0.0.1:{server.port}")
+        with pytest.raises(grpc.RpcError) as exc:
+            client.call("S", "boom", {})
+        assert exc.value.code() == grpc.StatusCode.INTERNAL
+    finally:
+        server.stop(0)
+
+
+def test_master_client_degrades_when_master_gone():
+    from elasticdl_amd.worker.master_client import MasterClient
+
+    mc = MasterClient("127.0.0.1:1", worker_id=0)  # nothing listening
+    task = mc.get_task()
+    assert task.type == TaskType.NONE
+    mc.report_task_result(1)  # must not raise
+    info = mc.get_comm_rank()
+    assert info["rank_id"] == -1
+    assert mc.job_finished()
+
+
+def test_worker_wait_then_tasks_appear():
+    from elasticdl_amd.master.servicer import MasterServicer
+    from elasticdl_amd.master.task_manager import TaskManager
+    from elasticdl_amd.utils.model_utils import get_model_spec
+    from elasticdl_amd.worker.master_client import MasterClient
+    from elasticdl_amd.worker.trainer import LocalTrainer
+    from elasticdl_amd.worker.worker import Worker
+
+    spec = get_model_spec("mnist")
+    reader = spec.data_reader_fn("synthetic:32")
+    tm = TaskManager()  # worker-driven: worker WAITs until params arrive
+    servicer = MasterServicer(tm)
+    server = rpc.start_server("127.0.0.1:0", {"Master": servicer.methods()})
+    try:
+        mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=0)
+        worker = Worker(0, mc, LocalTrainer(spec, "cpu"), data_reader=reader,
+                        spec=spec, minibatch_size=16)
+        t = threading.Thread(target=worker.run)
+        t.start()
+        time.sleep(1.0)  # worker should be WAITing
+        assert not tm.finished()
+        mc.report_training_params(dataset_size=32, batch_size=16)
+        t.join(60)
+        assert not t.is_alive()
+        assert tm.finished()
+    finally:
+        server.stop(0)
