@@ -164,8 +164,14 @@ class Worker:
         self.mc.report_task_result(task.task_id)
 
     def _run_prediction_task(self, task: Task) -> None:
+        # optional zoo hook consumes prediction outputs (the reference
+        # routes them through prediction-outputs processors)
+        process = getattr(self.spec.module, "process_predictions", None) \
+            if self.spec else None
         for batch in self._minibatches(task):
-            self.trainer.predict_minibatch(batch)
+            out = self.trainer.predict_minibatch(batch)
+            if process is not None:
+                process(out)
         self.mc.report_task_result(task.task_id)
 
     def _run_train_end_task(self, task: Task) -> None:

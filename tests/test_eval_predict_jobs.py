@@ -43,11 +43,17 @@ def test_evaluate_only_job():
 def test_predict_only_job():
     spec = get_model_spec("mnist")
     reader = spec.data_reader_fn("synthetic:64")
-    tm = TaskManager(prediction_shards=[("mnist-synthetic", 0, 64)],
-                     records_per_task=32)
-    assert tm.create_prediction_tasks() == 2
-    run_job(tm, spec, reader, eval_reader=reader)
-    assert tm.finished()
+    collected = []
+    spec.module.process_predictions = collected.append  # zoo output hook
+    try:
+        tm = TaskManager(prediction_shards=[("mnist-synthetic", 0, 64)],
+                         records_per_task=32)
+        assert tm.create_prediction_tasks() == 2
+        run_job(tm, spec, reader, eval_reader=reader)
+        assert tm.finished()
+        assert sum(t.shape[0] for t in collected) == 64
+    finally:
+        del spec.module.process_predictions
 
 
 def test_record_index_service():
