@@ -90,7 +90,7 @@ class MasterServicer:
             and task.type == TaskType.EVALUATION
             and self._evaluation_service is not None
         ):
-            if self._task_manager._eval_todo_count == 0:
+            if self._task_manager.pending_evaluation_tasks == 0:
                 self._evaluation_service.complete_task()
         return {}
 

@@ -371,6 +371,10 @@ class TaskManager:
             return True
 
     @property
+    def pending_evaluation_tasks(self) -> int:
+        return self._eval_todo_count
+
+    @property
     def completed_steps(self) -> int:
         return self._completed_steps
 
