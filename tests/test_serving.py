@@ -39,3 +39,14 @@ def test_serve_exported_mnist_model():
         bad = client.post("/v1/models/default:predict",
                           json={"instances": [["oops"]]})
         assert bad.status_code == 400
+
+        # binary path: codec body in, codec body out, same numerics
+        from elasticdl_amd.common import codec
+
+        rb = client.post(
+            "/v1/models/default:predict_binary",
+            content=codec.encode({"instances": x}),
+        )
+        assert rb.status_code == 200, rb.text
+        out = codec.decode(rb.content)["predictions"]
+        assert torch.allclose(out, ref, atol=1e-5)
