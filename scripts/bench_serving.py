@@ -14,23 +14,36 @@ from elasticdl_amd.serving.server import ModelRunner  # noqa: E402
 
 
 def bench(runner, make_batch, sizes, iters=50, warmup=10):
+    from elasticdl_amd.common import codec
+
+    name = runner.spec.module.__name__.split(".")[-1]
     for bs in sizes:
-        x = make_batch(bs).tolist()
-        for _ in range(warmup):
-            runner.predict(x)
-        if torch.cuda.is_available():
-            torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        for _ in range(iters):
-            runner.predict(x)
-        if torch.cuda.is_available():
-            torch.cuda.synchronize()
-        dt = (time.perf_counter() - t0) / iters
-        print(json.dumps({
-            "bench": f"serve_{runner.spec.module.__name__.split('.')[-1]}_bs{bs}",
-            "p50_ms": round(dt * 1e3, 3),
-            "qps": round(bs / dt, 1),
-        }), flush=True)
+        xt = make_batch(bs)
+        for mode in ("json", "binary"):
+            if mode == "json":
+                payload = xt.tolist()
+                call = lambda: runner.predict(payload)
+            else:
+                body = codec.encode({"instances": xt})
+                call = lambda: codec.encode(
+                    {"predictions": runner.predict_tensor(
+                        codec.decode(body)["instances"])}
+                )
+            for _ in range(warmup):
+                call()
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(iters):
+                call()
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / iters
+            print(json.dumps({
+                "bench": f"serve_{name}_bs{bs}_{mode}",
+                "p50_ms": round(dt * 1e3, 3),
+                "qps": round(bs / dt, 1),
+            }), flush=True)
 
 
 def main():
