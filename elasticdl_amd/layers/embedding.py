@@ -27,7 +27,10 @@ class _PsLookup(torch.autograd.Function):
             rows = module.lookup_fn(module.name, flat_ids)
         ctx.module = module
         ctx.ids = flat_ids
-        return rows.to(trigger.dtype) if trigger.dtype != rows.dtype else rows
+        # rows stay in PS precision (f32); the module applies ONE cast to
+        # its out_dtype afterwards (a trigger-dtype intermediate cast
+        # would round-trip f32->bf16->f32 and lose mantissa bits)
+        return rows
 
     @staticmethod
     def backward(ctx, grad_rows):
