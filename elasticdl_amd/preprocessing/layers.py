@@ -47,13 +47,21 @@ class Hashing(nn.Module):
         digest = hashlib.md5(str(s).encode("utf-8")).hexdigest()[:16]
         return int(digest, 16) % self.num_bins
 
+    @staticmethod
+    def _splitmix64(x: torch.Tensor) -> torch.Tensor:
+        # same avalanche finalizer as the GPU hash table (ps_kernels.hip);
+        # int64 wrap-around == uint64 arithmetic mod 2^64
+        x = x + 0x9E3779B97F4A7C15
+        x = (x ^ (x >> 30)) * -0x40A7B892E31B1A47  # 0xBF58476D1CE4E5B9
+        x = (x ^ (x >> 27)) * -0x6B2FB644ECCEEE15  # 0x94D049BB133111EB
+        return x ^ (x >> 31)
+
     def forward(self, inputs):
         if isinstance(inputs, torch.Tensor):
+            # vectorized integer path (strings go through md5 below)
             def hash_tensor(t):
-                return torch.as_tensor(
-                    [self._hash_str(v) for v in t.reshape(-1).tolist()],
-                    dtype=torch.int64,
-                ).reshape(t.shape)
+                h = self._splitmix64(t.to(torch.int64))
+                return h.remainder(self.num_bins)
 
             return _map_values(inputs.to(torch.int64), hash_tensor)
         if isinstance(inputs, (list, tuple)):
