@@ -29,6 +29,14 @@ def build_parser() -> argparse.ArgumentParser:
         p = sub.add_parser(name)
         add_common_train_params(p)
         add_k8s_params(p)
+
+    sv = sub.add_parser("serve", help="serve an exported model over HTTP")
+    sv.add_argument("--model_def", required=True)
+    sv.add_argument("--model_path", default="")
+    sv.add_argument("--model_params", default="")
+    sv.add_argument("--device", default="auto")
+    sv.add_argument("--host", default="0.0.0.0")
+    sv.add_argument("--port", type=int, default=8500)
     return parser
 
 
@@ -44,6 +52,14 @@ def main(argv=None) -> int:
         if args.zoo_command == "push":
             return api.push_zoo(args.image)
         parser.error("zoo subcommand required")
+    if args.command == "serve":
+        from elasticdl_amd.serving.server import main as serve_main
+
+        return serve_main([
+            "--model_def", args.model_def, "--model_path", args.model_path,
+            "--model_params", args.model_params, "--device", args.device,
+            "--host", args.host, "--port", str(args.port),
+        ])
     if args.command in ("train", "evaluate", "predict"):
         return api.submit_job(args, args.command)
     parser.error("command required")

@@ -75,6 +75,10 @@ class TaskManager:
 
         self.total_records = sum(s[2] - s[1] for s in self._training_shards)
         self.failed_records = 0
+        self._completed_records = 0
+        # (timestamp, completed_records) samples for throughput reporting
+        # (per-step counters for the BASELINE curves)
+        self._throughput_log: List[Tuple[float, int]] = []
 
         # worker-driven jobs (SDK path): no shards at startup — the job is
         # not "finished" until a worker reports training params or tasks
@@ -282,6 +286,13 @@ class TaskManager:
                 )
                 if task.type == TaskType.TRAINING:
                     self._completed_steps += 1
+                    if task.shard is not None:
+                        self._completed_records += task.shard.size
+                        self._throughput_log.append(
+                            (time.monotonic(), self._completed_records)
+                        )
+                        if len(self._throughput_log) > 256:
+                            self._throughput_log = self._throughput_log[-128:]
                 elif task.type == TaskType.EVALUATION:
                     self._eval_todo_count = max(0, self._eval_todo_count - 1)
                 elif task.type == TaskType.TRAIN_END_CALLBACK:
@@ -384,10 +395,17 @@ class TaskManager:
 
     def counts(self) -> Dict[str, int]:
         with self._lock:
+            rate = 0.0
+            if len(self._throughput_log) >= 2:
+                (t0, r0), (t1, r1) = self._throughput_log[0], self._throughput_log[-1]
+                if t1 > t0:
+                    rate = (r1 - r0) / (t1 - t0)
             return {
                 "todo": len(self._todo),
                 "doing": len(self._doing),
                 "completed_steps": self._completed_steps,
+                "completed_records": self._completed_records,
+                "records_per_sec": round(rate, 2),
                 "epoch": self._epoch,
                 "failed_records": self.failed_records,
             }
