@@ -117,3 +117,27 @@ def test_worker_wait_then_tasks_appear():
         assert tm.finished()
     finally:
         server.stop(0)
+
+
+def test_cli_serve_parser():
+    from elasticdl_amd.client.main import build_parser
+
+    args = build_parser().parse_args([
+        "serve", "--model_def", "mnist", "--port", "9000",
+    ])
+    assert args.command == "serve"
+    assert args.port == 9000
+
+
+def test_task_manager_throughput_counters():
+    from elasticdl_amd.master.task_manager import TaskManager
+
+    tm = TaskManager(training_shards=[("f", 0, 64)], records_per_task=16)
+    while True:
+        t = tm.get(0)
+        if t.type != TaskType.TRAINING:
+            break
+        tm.report(t.task_id, True, 0)
+    c = tm.counts()
+    assert c["completed_records"] == 64
+    assert c["records_per_sec"] >= 0
