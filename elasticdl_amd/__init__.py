@@ -13,3 +13,27 @@ A from-scratch rebuild of the capabilities of sql-machine-learning/elasticdl
 """
 
 __version__ = "0.1.0"
+
+
+def __getattr__(name):
+    """Lazy top-level conveniences (torch import stays off the module
+    import path until needed)."""
+    _exports = {
+        "EdlEmbedding": ("elasticdl_amd.layers.embedding", "EdlEmbedding"),
+        "FusedDense": ("elasticdl_amd.ops.functional", "FusedDense"),
+        "DistributedOptimizer": (
+            "elasticdl_amd.collective.distributed_optimizer",
+            "DistributedOptimizer",
+        ),
+        "ElasticAllReduceController": (
+            "elasticdl_amd.collective.controller",
+            "ElasticAllReduceController",
+        ),
+        "PSEngine": ("elasticdl_amd.ps.engine", "PSEngine"),
+    }
+    if name in _exports:
+        import importlib
+
+        mod, attr = _exports[name]
+        return getattr(importlib.import_module(mod), attr)
+    raise AttributeError(name)

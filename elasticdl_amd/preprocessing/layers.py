@@ -34,8 +34,10 @@ def _map_values(inputs: torch.Tensor, fn):
 
 
 class Hashing(nn.Module):
-    """Deterministic string/int hash to [0, num_bins)
-    (reference: hashing.py:19 — tf.strings.to_hash_bucket_fast)."""
+    """Deterministic hash to [0, num_bins)
+    (reference: hashing.py:19 — tf.strings.to_hash_bucket_fast).
+    Integer tensors use a vectorized splitmix64 (matching the GPU hash
+    table's mixer); strings hash via md5 (CPU feature engineering)."""
 
     def __init__(self, num_bins: int):
         super().__init__()
