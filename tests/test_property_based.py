@@ -1,0 +1,76 @@
+"""Property-based checks (hypothesis) for the codec and dedup semantics."""
+
+import hypothesis.strategies as st
+import torch
+from hypothesis import given, settings
+
+from elasticdl_amd.common import codec
+from elasticdl_amd.common.tensor_utils import deduplicate_indexed_slices
+
+scalars = st.one_of(
+    st.integers(min_value=-(2**62), max_value=2**62),
+    st.floats(allow_nan=False, allow_infinity=False, width=64),
+    st.text(max_size=20),
+    st.booleans(),
+    st.none(),
+)
+
+structures = st.recursive(
+    scalars,
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(st.text(max_size=8), children, max_size=4),
+    ),
+    max_leaves=12,
+)
+
+
+@settings(max_examples=60, deadline=None)
+@given(structures)
+def test_codec_roundtrip_structures(msg):
+    assert codec.decode(codec.encode(msg)) == msg
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    st.lists(st.integers(min_value=0, max_value=30), min_size=1, max_size=64),
+    st.integers(min_value=1, max_value=6),
+)
+def test_dedup_matches_naive(ids, dim):
+    ids_t = torch.tensor(ids, dtype=torch.int64)
+    vals = torch.randn(len(ids), dim)
+    summed, uids = deduplicate_indexed_slices(vals, ids_t)
+    # naive oracle
+    expect = {}
+    for i, v in enumerate(ids):
+        expect[v] = expect.get(v, torch.zeros(dim)) + vals[i]
+    assert uids.tolist() == sorted(set(ids))
+    for j, v in enumerate(uids.tolist()):
+        assert torch.allclose(summed[j], expect[v], atol=1e-5)
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    st.lists(
+        st.tuples(st.text(max_size=6), st.integers(0, 4), st.integers(5, 9)),
+        min_size=0, max_size=5,
+    ),
+    st.integers(min_value=1, max_value=7),
+)
+def test_task_shard_math_covers_all_records(shards, rpt):
+    from elasticdl_amd.common.task import TaskType
+    from elasticdl_amd.master.task_manager import TaskManager
+
+    tm = TaskManager(
+        training_shards=[(n, a, b) for n, a, b in shards],
+        records_per_task=rpt,
+    )
+    total = 0
+    while True:
+        t = tm.get(0)
+        if t.type != TaskType.TRAINING:
+            break
+        assert 0 < t.shard.size <= rpt
+        total += t.shard.size
+        tm.report(t.task_id, True, 0)
+    assert total == sum(b - a for _, a, b in shards)
