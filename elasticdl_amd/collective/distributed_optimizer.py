@@ -272,6 +272,40 @@ class DistributedOptimizer:
     def set_backward_passes_per_step(self, n: int) -> None:
         self.backward_passes_per_step = max(1, n)
 
+    def ensure_state(self) -> None:
+        """Materialize all optimizer slot tensors (vel / m,v) eagerly.
+
+        Lazy allocation is fine locally, but a worker that joins after an
+        elasticity event must receive rank 0's momentum/Adam state (the
+        reference relies on Horovod's broadcast_optimizer_state,
+        elasticai_api/pytorch/controller.py:126-131); eager allocation
+        guarantees every rank has the same state tensors to broadcast into.
+        """
+        for b in self.buckets:
+            if self._native:
+                if self.opt_type == "sgd":
+                    b.state.setdefault("vel", torch.zeros_like(b.master))
+                else:
+                    b.state.setdefault("m", torch.zeros_like(b.master))
+                    b.state.setdefault("v", torch.zeros_like(b.master))
+            else:
+                for idx, p in enumerate(b.params):
+                    zero = lambda: torch.zeros(
+                        p.shape, dtype=torch.float32, device=p.device
+                    )
+                    if self.opt_type == "sgd":
+                        b.state.setdefault(f"vel{idx}", zero())
+                    else:
+                        b.state.setdefault(f"m{idx}", zero())
+                        b.state.setdefault(f"v{idx}", zero())
+
+    def set_step_count(self, n: int) -> None:
+        self._step_count = int(n)
+
+    @property
+    def step_count(self) -> int:
+        return self._step_count
+
     def state_dict(self) -> dict:
         return {
             "step": self._step_count,

@@ -99,9 +99,13 @@ def parse_model_params(s: str) -> dict:
         if not part:
             continue
         k, _, v = part.partition("=")
+        # literal_eval only: model_params flows in from CLI/k8s job specs,
+        # so arbitrary-expression evaluation is off the table
         try:
-            out[k.strip()] = eval(v, {"__builtins__": {}})  # noqa: S307 - literals
-        except Exception:  # noqa: BLE001
+            import ast
+
+            out[k.strip()] = ast.literal_eval(v.strip())
+        except (ValueError, SyntaxError):
             out[k.strip()] = v.strip()
     return out
 

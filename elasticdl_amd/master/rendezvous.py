@@ -43,9 +43,14 @@ class ElasticRendezvousServer:
         self._ready_hosts = set()
         self._dead_hosts = set()
         self._cur_completed = True
-        # flip debounce: after staging, wait a moment so racing removals of
-        # the same elasticity event coalesce into one generation
+        # flip debounce: a staged world only flips once it has been stable
+        # for this long, so racing removals of the same elasticity event
+        # coalesce into one generation. Implemented as a deadline (staged_at
+        # timestamp), NOT a sleep: sleeping under self._lock would serialize
+        # every get_comm_rank poll AND prevent the very membership calls the
+        # debounce is meant to observe from landing during the window.
         self._flip_delay_sec = 0.5
+        self._staged_at: Optional[float] = None
 
     # ------------------------------------------------------------- lifecycle
     def start(self) -> int:
@@ -87,6 +92,7 @@ class ElasticRendezvousServer:
             # total wipeout, so the refusal is intentionally dropped.
             if worker_host not in self._next_hosts:
                 self._next_hosts.append(worker_host)
+                self._staged_at = time.monotonic()
                 logger.info(
                     "Rendezvous: staged add of %s (next world %s)",
                     worker_host,
@@ -107,6 +113,7 @@ class ElasticRendezvousServer:
                 # deadlocked (improves on the reference, which can stall if
                 # a worker dies between world formation and completion)
                 self._dead_hosts.add(worker_host)
+                self._staged_at = time.monotonic()
                 logger.info(
                     "Rendezvous: staged removal of %s (next world %s)",
                     worker_host,
@@ -121,6 +128,7 @@ class ElasticRendezvousServer:
         with self._lock:
             if self._next_hosts is None:
                 self._next_hosts = copy.deepcopy(self._cur_hosts)
+            self._staged_at = time.monotonic()
             logger.info("Rendezvous: reset requested (next world %s)",
                         self._next_hosts)
 
@@ -129,8 +137,15 @@ class ElasticRendezvousServer:
         """One-stop poll for workers: rank/world/rendezvous_id/store addr."""
         with self._lock:
             self._mark_ready(worker_host)
-            if self._next_hosts is not None and self._cur_completed:
-                time.sleep(self._flip_delay_sec)
+            if (
+                self._next_hosts is not None
+                and self._cur_completed
+                and (
+                    self._staged_at is None
+                    or time.monotonic() - self._staged_at
+                    >= self._flip_delay_sec
+                )
+            ):
                 self._flip()
                 self._mark_ready(worker_host)
             rank = (
@@ -161,6 +176,7 @@ class ElasticRendezvousServer:
     def _flip(self) -> None:
         self._cur_hosts = self._next_hosts
         self._next_hosts = None
+        self._staged_at = None
         self._rendezvous_id += 1
         # an empty world can never report readiness — it is trivially
         # complete, so the next staged world can flip immediately

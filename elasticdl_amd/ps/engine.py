@@ -67,6 +67,7 @@ class PSEngine:
         self._pending_dense: Dict[str, torch.Tensor] = {}
         self._pending_sparse: Dict[str, List[IndexedSlices]] = {}
         self._pending_count = 0
+        self._pending_lr: Optional[float] = None
 
         self.version_listeners = []  # callables(version) after each update
 
@@ -103,14 +104,22 @@ class PSEngine:
                     self._create_table(info)
 
     def _create_table(self, info: dict) -> None:
+        import hashlib
+
         init = info.get("initializer", ["uniform", -0.05, 0.05])
+        # per-table seed must be deterministic across processes/restarts
+        # (Python's hash() is randomized per process via PYTHONHASHSEED,
+        # which would break the id-keyed-init reproducibility guarantee)
+        name_seed = int.from_bytes(
+            hashlib.sha256(info["name"].encode()).digest()[:4], "little"
+        )
         self.tables[info["name"]] = EmbeddingTable(
             name=info["name"],
             dim=info["dim"],
             device=self.device,
             max_rows=info.get("max_rows", self.embedding_max_rows),
             initializer=(init[0], float(init[1]), float(init[2])),
-            seed=self.seed + hash(info["name"]) % (1 << 31),
+            seed=(self.seed + name_seed) % (1 << 31),
         )
 
     # ---------------------------------------------------------------- pulls
@@ -137,30 +146,46 @@ class PSEngine:
         self,
         dense_grads: Dict[str, torch.Tensor],
         embedding_grads: Dict[str, IndexedSlices],
-        learning_rate: float = 1.0,
+        learning_rate: Optional[float] = None,
         version: int = 0,
     ) -> Tuple[bool, int]:
-        """Returns (accepted, current_version)."""
-        if self.use_async:
-            return self._push_async(dense_grads, embedding_grads, version)
-        return self._push_sync(dense_grads, embedding_grads, version)
+        """Returns (accepted, current_version).
 
-    def _lr_mult_for(self, grad_version: int) -> float:
+        `learning_rate` is the worker-supplied LR carried in PushGradients
+        (go/pkg/ps/server.go:176-206; the version-keyed LearningRateScheduler
+        callback computes it worker-side, elasticdl/callbacks.py:69-109).
+        When given it REPLACES the optimizer's base LR for this update;
+        staleness modulation still multiplies on top.  None keeps base LR.
+        """
+        if self.use_async:
+            return self._push_async(
+                dense_grads, embedding_grads, version, learning_rate
+            )
+        return self._push_sync(
+            dense_grads, embedding_grads, version, learning_rate
+        )
+
+    def _lr_mult_for(self, grad_version: int, learning_rate=None) -> float:
+        mult = 1.0
+        if learning_rate is not None and self.optimizer.base_lr > 0:
+            mult = float(learning_rate) / self.optimizer.base_lr
         if self.lr_staleness_modulation:
             staleness = max(1, self.version - grad_version)
-            return 1.0 / staleness
-        return 1.0
+            mult /= staleness
+        return mult
 
-    def _push_async(self, dense_grads, embedding_grads, version) -> Tuple[bool, int]:
+    def _push_async(self, dense_grads, embedding_grads, version,
+                    learning_rate=None) -> Tuple[bool, int]:
         with self._lock:
-            lr_mult = self._lr_mult_for(version)
+            lr_mult = self._lr_mult_for(version, learning_rate)
             self._apply(dense_grads, embedding_grads, lr_mult)
             self.version += 1
             v = self.version
         self._notify(v)
         return True, v
 
-    def _push_sync(self, dense_grads, embedding_grads, version) -> Tuple[bool, int]:
+    def _push_sync(self, dense_grads, embedding_grads, version,
+                   learning_rate=None) -> Tuple[bool, int]:
         with self._lock:
             if version < self.version - self.sync_version_tolerance:
                 return False, self.version  # stale, worker must re-pull
@@ -178,6 +203,8 @@ class PSEngine:
                     )
                 )
             self._pending_count += 1
+            if learning_rate is not None:
+                self._pending_lr = float(learning_rate)
             if self._pending_count < self.grads_to_wait:
                 return True, self.version
             # averaged dense / merged (summed) sparse, single apply
@@ -189,7 +216,11 @@ class PSEngine:
                 name: merge_indexed_slices(*lst)
                 for name, lst in self._pending_sparse.items()
             }
-            self._apply(dense_avg, sparse_merged, 1.0)
+            self._apply(
+                dense_avg, sparse_merged,
+                self._lr_mult_for(self.version, self._pending_lr),
+            )
+            self._pending_lr = None
             self._pending_dense.clear()
             self._pending_sparse.clear()
             self._pending_count = 0

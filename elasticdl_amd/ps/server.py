@@ -83,7 +83,7 @@ class PserverServicer:
         accepted, version = self.engine.push_gradients(
             req.get("dense_gradients", {}),
             embedding,
-            learning_rate=req.get("learning_rate", 1.0),
+            learning_rate=req.get("learning_rate"),
             version=req.get("version", 0),
         )
         return {"accepted": accepted, "version": version}

@@ -29,14 +29,29 @@ class SavedModelExporter(Callback):
 class LearningRateScheduler(Callback):
     """``multiplier_fn(model_version) -> float`` scales the base LR
     (reference: callbacks.py:69-109, LR keyed on model version so async
-    workers agree on the schedule)."""
+    workers agree on the schedule).
 
-    def __init__(self, optimizer, multiplier_fn: Callable[[int], float]):
+    Two forms:
+    - ``LearningRateScheduler(optimizer, fn)`` — adjusts a local/allreduce
+      optimizer's LR before each batch;
+    - ``LearningRateScheduler(fn)`` — no optimizer; the PS-strategy trainer
+      reads ``multiplier_fn`` and ships base_lr*mult(version) to the PS in
+      each PushGradients (the reference's PS-path LR scheduling,
+      go/pkg/ps/server.go:176-206).
+    """
+
+    def __init__(self, optimizer, multiplier_fn: Callable[[int], float] = None):
+        if multiplier_fn is None and callable(optimizer) \
+                and not hasattr(optimizer, "param_groups") \
+                and not hasattr(optimizer, "lr"):
+            optimizer, multiplier_fn = None, optimizer
         self.optimizer = optimizer
         self.multiplier_fn = multiplier_fn
         self._base_lrs = None
 
     def on_train_batch_begin(self, version: int) -> None:
+        if self.optimizer is None:
+            return
         mult = self.multiplier_fn(max(0, version))
         if hasattr(self.optimizer, "param_groups"):  # torch optimizer
             if self._base_lrs is None:

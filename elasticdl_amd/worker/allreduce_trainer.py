@@ -63,6 +63,7 @@ class AllReduceTrainer(Trainer):
         self._completed_batches = 0
         self._last_world_check = 0.0
         self._version = 0
+        self._need_zero = True
 
     # ---------------------------------------------------------- elasticity
     def init_communicator_if_needed(self, force: bool = False) -> None:
@@ -87,6 +88,10 @@ class AllReduceTrainer(Trainer):
             self.comm.world_size,
             self.comm.rendezvous_id,
         )
+        # Slot tensors must exist on every rank before broadcasting them —
+        # a rejoining worker would otherwise restart with zero momentum and
+        # silently diverge from the cohort (each rank applies its own step).
+        self.opt.ensure_state()
         for b in self.opt.buckets:
             if b.param_flat is not None:
                 dist.broadcast(b.param_flat, 0)
@@ -94,11 +99,16 @@ class AllReduceTrainer(Trainer):
             else:
                 for p in b.params:
                     dist.broadcast(p.data, 0)
+            for k in sorted(b.state):
+                dist.broadcast(b.state[k], 0)
         for t in self.model.buffers():
             if t.numel() > 0 and t.dtype.is_floating_point:
                 dist.broadcast(t.data, 0)
         self._completed_batches = int(
             self.comm.broadcast_value(float(self._completed_batches), 0)
+        )
+        self.opt.set_step_count(
+            int(self.comm.broadcast_value(float(self.opt.step_count), 0))
         )
 
     def _adjust_accumulation(self) -> None:
@@ -125,6 +135,11 @@ class AllReduceTrainer(Trainer):
                         e, attempt + 1, MAX_ALLREDUCE_RETRY_NUM,
                     )
                     self.comm.handle_collective_failure()
+                    # partial accumulation is unusable after a world change
+                    # (the per-rank micro-batch split changed mid-step):
+                    # drop it and start the step over, like the reference's
+                    # restore() path (pytorch/controller.py:133-164)
+                    self._need_zero = True
                     time.sleep(1)
                     continue
                 raise
@@ -132,7 +147,12 @@ class AllReduceTrainer(Trainer):
 
     def _train_once(self, batch):
         x, y = self._feed(batch)
-        self.opt.zero_grad()
+        # zero only at step boundaries: zero_grad() resets the accumulation
+        # counter AND the bucket buffers, so calling it per micro-batch
+        # would silently disable backward_passes_per_step > 1 accumulation
+        if self._need_zero:
+            self.opt.zero_grad()
+            self._need_zero = False
         out = self.model(x)
         loss = self.spec.loss_fn(out.float(), y)
         loss.backward()
@@ -140,6 +160,7 @@ class AllReduceTrainer(Trainer):
         if step_due:
             self.opt.step()
             self._version += 1
+            self._need_zero = True
         self._completed_batches += 1
         return loss.detach(), self._version
 

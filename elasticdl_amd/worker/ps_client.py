@@ -105,7 +105,7 @@ class PSClient:
         self,
         dense_grads: Dict[str, torch.Tensor],
         edl_grads: Dict[str, List[IndexedSlices]] = None,
-        learning_rate: float = 1.0,
+        learning_rate: Optional[float] = None,
         version: int = 0,
     ) -> (bool, int):
         """Dense grads go to their name shard; embedding grads are merged,
@@ -135,18 +135,14 @@ class PSClient:
         for s, c in enumerate(self._clients):
             if not per_shard_dense[s] and not per_shard_emb[s]:
                 continue
-            futs.append(
-                c.call_future(
-                    "Pserver",
-                    "push_gradients",
-                    {
-                        "dense_gradients": per_shard_dense[s],
-                        "embedding_gradients": per_shard_emb[s],
-                        "learning_rate": learning_rate,
-                        "version": version,
-                    },
-                )
-            )
+            msg = {
+                "dense_gradients": per_shard_dense[s],
+                "embedding_gradients": per_shard_emb[s],
+                "version": version,
+            }
+            if learning_rate is not None:
+                msg["learning_rate"] = float(learning_rate)
+            futs.append(c.call_future("Pserver", "push_gradients", msg))
         accepted = True
         max_version = 0
         for f in futs:

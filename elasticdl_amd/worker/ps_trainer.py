@@ -65,6 +65,27 @@ class ParameterServerTrainer(Trainer):
         if self.get_model_steps > 1:
             self._local_opt = self._build_local_optimizer()
 
+        # version-keyed LR scheduling on the PS path: the worker computes
+        # lr = base_lr * mult(version) and carries it in PushGradients
+        # (reference callbacks.py:69-109 + go/pkg/ps/server.go:176-206)
+        from elasticdl_amd.ps.optimizer import parse_opt_args
+        from elasticdl_amd.utils.callbacks import LearningRateScheduler
+
+        self._base_lr = float(
+            parse_opt_args(self.opt_args).get("learning_rate", 0.01)
+        )
+        self._lr_mult_fn = None
+        if spec.callbacks_fn is not None:
+            for cb in spec.callbacks_fn() or []:
+                if isinstance(cb, LearningRateScheduler) and cb.multiplier_fn:
+                    self._lr_mult_fn = cb.multiplier_fn
+
+    def current_learning_rate(self):
+        """Scheduled LR for this step, or None to use the PS base LR."""
+        if self._lr_mult_fn is None:
+            return None
+        return self._base_lr * float(self._lr_mult_fn(max(0, self._version)))
+
     def _build_local_optimizer(self):
         from elasticdl_amd.ps.optimizer import parse_opt_args
 
@@ -129,9 +150,13 @@ class ParameterServerTrainer(Trainer):
         edl_grads: Dict[str, List] = {}
         for name, slices in self._grad_sink:
             edl_grads.setdefault(name, []).append(slices)
+        lr = self.current_learning_rate()
+        if lr is not None and self._local_opt is not None:
+            for g in self._local_opt.param_groups:
+                g["lr"] = lr
         self.timing.start_record_time("report_gradient")
         accepted, version = self.ps.push_gradients(
-            dense_grads, edl_grads, version=self._version
+            dense_grads, edl_grads, learning_rate=lr, version=self._version
         )
         self.timing.end_record_time("report_gradient")
         if not accepted:
