@@ -22,6 +22,26 @@ ELASTICDL_REPLICA_TYPE_KEY = "elasticdl-replica-type"
 ELASTICDL_REPLICA_INDEX_KEY = "elasticdl-replica-index"
 
 
+def k8s_types():
+    """The ``kubernetes.client`` module when installed, else the offline
+    attribute-bag equivalents (k8s_types.py) so spec building stays
+    testable without the cluster SDK."""
+    try:
+        from kubernetes import client as k8s
+
+        return k8s
+    except ImportError:
+        from elasticdl_amd.master import k8s_types as k8s
+
+        return k8s
+
+# fixed in-pod service ports (reference: common/k8s_client.py:29-30) —
+# every PS pod serves on the same port behind its own per-pod Service,
+# so cross-pod addresses are stable DNS names independent of pod IPs
+PS_SERVICE_PORT = 2222
+WORKER_SERVICE_PORT = 3333
+
+
 def parse_resource(spec: str) -> Dict[str, str]:
     """'cpu=4,memory=8192Mi,amd.com/gpu=1' -> k8s resource dict
     (reference: elasticdl_client/common/k8s_resource.py)."""
@@ -72,6 +92,11 @@ class ClusterSpec:
             return self._mod.patch_pod(pod, pod_type) or pod
         return pod
 
+    def patch_service(self, service):
+        if self._mod is not None and hasattr(self._mod, "patch_service"):
+            return self._mod.patch_service(service) or service
+        return service
+
 
 class Client:
     def __init__(
@@ -110,6 +135,20 @@ class Client:
     def get_pod_name(self, pod_type: str, index: int) -> str:
         return f"elasticdl-{self.job_name}-{pod_type}-{index}"
 
+    def get_service_name(self, pod_type: str, index: int) -> str:
+        """Service name == pod name (reference: get_ps_service_name)."""
+        return self.get_pod_name(pod_type, index)
+
+    def get_service_address(self, pod_type: str, index: int,
+                            port: int) -> str:
+        """Cluster-DNS address of a replica's Service
+        (reference: common/k8s_client.py:113-114)."""
+        return f"{self.get_service_name(pod_type, index)}." \
+               f"{self.namespace}.svc:{port}"
+
+    def get_ps_service_address(self, ps_id: int) -> str:
+        return self.get_service_address("ps", ps_id, PS_SERVICE_PORT)
+
     # --------------------------------------------------------------- CRUD
     def create_pod(self, pod) -> bool:
         try:
@@ -121,16 +160,11 @@ class Client:
 
     def delete_pod(self, pod_name: str) -> bool:
         try:
-            from kubernetes import client as k8s
-
             self.client.delete_namespaced_pod(
                 pod_name,
                 self.namespace,
-                body=k8s.V1DeleteOptions(grace_period_seconds=0),
+                body=k8s_types().V1DeleteOptions(grace_period_seconds=0),
             )
-            return True
-        except ImportError:
-            self.client.delete_namespaced_pod(pod_name, self.namespace)
             return True
         except Exception:  # noqa: BLE001
             logger.warning("delete_pod(%s) failed", pod_name)
@@ -141,6 +175,89 @@ class Client:
             return self.client.read_namespaced_pod(pod_name, self.namespace)
         except Exception:  # noqa: BLE001
             return None
+
+    def create_service(self, service) -> bool:
+        try:
+            self.client.create_namespaced_service(self.namespace, service)
+            return True
+        except Exception:  # noqa: BLE001
+            logger.warning("create_service failed:\n%s",
+                           traceback.format_exc())
+            return False
+
+    def patch_service(self, service_name: str, service) -> bool:
+        try:
+            self.client.patch_namespaced_service(
+                service_name, self.namespace, service
+            )
+            return True
+        except Exception:  # noqa: BLE001
+            logger.warning("patch_service(%s) failed", service_name)
+            return False
+
+    def get_service(self, service_name: str):
+        try:
+            return self.client.read_namespaced_service(
+                service_name, self.namespace
+            )
+        except Exception:  # noqa: BLE001
+            return None
+
+    def build_service_spec(
+        self,
+        pod_type: str,
+        index: int,
+        port: int,
+        target_port: int = 0,
+        owner_pod=None,
+        selector_index: Optional[int] = None,
+    ):
+        """V1Service selecting one replica by (job, type, index) labels
+        (reference: common/k8s_client.py:275-311). ``selector_index`` lets
+        a relaunched worker be patched behind the original service name."""
+        k8s = k8s_types()
+
+        labels = {
+            "app": "elasticdl",
+            ELASTICDL_JOB_KEY: self.job_name,
+            ELASTICDL_REPLICA_TYPE_KEY: pod_type,
+            ELASTICDL_REPLICA_INDEX_KEY: str(index),
+        }
+        selector = dict(labels)
+        if selector_index is not None:
+            selector[ELASTICDL_REPLICA_INDEX_KEY] = str(selector_index)
+        owner_refs = None
+        if owner_pod is not None:
+            owner_refs = [
+                k8s.V1OwnerReference(
+                    api_version="v1",
+                    kind="Pod",
+                    name=owner_pod.metadata.name,
+                    uid=owner_pod.metadata.uid,
+                    block_owner_deletion=True,
+                    controller=True,
+                )
+            ]
+        return k8s.V1Service(
+            api_version="v1",
+            kind="Service",
+            metadata=k8s.V1ObjectMeta(
+                name=self.get_service_name(pod_type, index),
+                labels=labels,
+                # at least one annotation so cluster-spec hooks can extend
+                annotations=dict(labels),
+                owner_references=owner_refs,
+                namespace=self.namespace,
+            ),
+            spec=k8s.V1ServiceSpec(
+                ports=[
+                    k8s.V1ServicePort(
+                        port=port, target_port=target_port or port
+                    )
+                ],
+                selector=selector,
+            ),
+        )
 
     def patch_labels_to_pod(self, pod_name: str, labels: Dict[str, str]):
         try:
@@ -169,7 +286,7 @@ class Client:
     ):
         """V1Pod with the elasticdl labels + owner reference to the master
         pod (reference: k8s_client.py:283-298)."""
-        from kubernetes import client as k8s
+        k8s = k8s_types()
 
         env = [k8s.V1EnvVar(name=k, value=str(v)) for k, v in (envs or {}).items()]
         env.append(
@@ -261,15 +378,25 @@ class Client:
         (reference: common/k8s_client.py:92-106) plus an optional periodic
         callback (retry pod creation etc.)."""
 
-        def watch_loop():
-            from kubernetes import watch
+        selector = f"{ELASTICDL_JOB_KEY}={self.job_name}"
 
+        def watch_loop():
             while True:
                 try:
+                    if hasattr(self.client, "stream_pod_events"):
+                        # fake/offline API: it owns the event stream
+                        # (returning ends the watch — no reconnect loop)
+                        for event in self.client.stream_pod_events(
+                            self.namespace, label_selector=selector
+                        ):
+                            event_callback(event)
+                        return
+                    from kubernetes import watch
+
                     stream = watch.Watch().stream(
                         self.client.list_namespaced_pod,
                         self.namespace,
-                        label_selector=f"{ELASTICDL_JOB_KEY}={self.job_name}",
+                        label_selector=selector,
                     )
                     for event in stream:
                         event_callback(event)

@@ -40,8 +40,9 @@ def _free_port() -> int:
 
 
 class Master:
-    def __init__(self, args):
+    def __init__(self, args, k8s_client=None):
         self.args = args
+        self._k8s_client = k8s_client  # test injection (fake CoreV1 API)
         self.stopped = threading.Event()
         self.exit_code = 0
 
@@ -119,7 +120,15 @@ class Master:
         )
 
         self.port = args.port or _free_port()
-        self.master_addr = f"127.0.0.1:{self.port}"
+        # In k8s mode the address handed to worker/PS pods must be routable
+        # from OTHER nodes: the master pod's IP (MY_POD_IP is injected into
+        # every elasticdl pod spec, reference elasticdl_job_service.py:65).
+        # Local mode keeps loopback.
+        if args.pod_manager == "k8s":
+            master_ip = os.environ.get("MY_POD_IP", "127.0.0.1")
+        else:
+            master_ip = "127.0.0.1"
+        self.master_addr = f"{master_ip}:{self.port}"
 
         # ---- pod manager
         self.pod_manager = None
@@ -215,9 +224,21 @@ class Master:
         return mgr
 
     def _create_k8s_pod_manager(self):
+        from elasticdl_amd.master.k8s_client import PS_SERVICE_PORT
         from elasticdl_amd.master.pod_manager import PodManager
 
-        mgr = PodManager(self.args, self)
+        mgr = PodManager(self.args, self, k8s_client=self._k8s_client)
+        # Every PS pod serves on the same fixed port behind its own
+        # per-pod Service; workers address PS shards by service DNS
+        # (reference pod_manager.py:269 + k8s_client.py:126-128). This is
+        # what makes cross-node PS discovery work — pod IPs are neither
+        # known at command-build time nor stable across relaunches.
+        self._ps_ports = [PS_SERVICE_PORT] * self.args.num_ps_pods
+        self.ps_addrs = [
+            mgr.k8s.get_ps_service_address(i)
+            for i in range(self.args.num_ps_pods)
+        ]
+        mgr.ps_addrs = self.ps_addrs
         mgr.add_pod_event_callback(TaskRescheduleCallback(self.task_manager))
         if self.rendezvous_server is not None:
             mgr.add_pod_event_callback(

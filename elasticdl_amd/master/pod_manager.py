@@ -83,12 +83,26 @@ class PodManager:
         self.k8s.start_watch(self._event_cb)
 
     def start_parameter_servers(self) -> None:
+        from elasticdl_amd.master.k8s_client import PS_SERVICE_PORT
+
         for i in range(self.args.num_ps_pods):
             self._start_pod(
                 "ps", i, self.master.ps_command(i),
                 self.args.ps_resource_request, self.args.ps_resource_limit,
                 self.args.ps_pod_priority,
             )
+            # one Service per PS pod: workers reach shard i at the stable
+            # DNS name elasticdl-<job>-ps-<i>.<ns>.svc:2222 (reference
+            # pod_manager.py:393-403 _start_ps + create_ps_service)
+            svc = self.k8s.build_service_spec(
+                pod_type="ps",
+                index=i,
+                port=PS_SERVICE_PORT,
+                owner_pod=self.k8s.get_pod(self.k8s.get_pod_name("ps", i)),
+            )
+            svc = self.cluster_spec.patch_service(svc)
+            if not self.k8s.create_service(svc):
+                logger.error("Failed to create PS service %d", i)
 
     def start_workers(self) -> None:
         for _ in range(self.args.num_workers):

@@ -29,6 +29,15 @@ POD_STATE_FLOW = [
     Transition(PodStatus.INITIAL, PodEventType.ADDED, "Pending", PodStatus.PENDING),
     Transition(PodStatus.INITIAL, PodEventType.ADDED, "Running", PodStatus.RUNNING),
     Transition(PodStatus.INITIAL, PodEventType.MODIFIED, "Pending", PodStatus.PENDING),
+    # a watch (re)connect can surface a pod first via MODIFIED/DELETED at
+    # any phase — INITIAL must accept them or the pod is tracked forever
+    # as INITIAL and never relaunched
+    Transition(PodStatus.INITIAL, PodEventType.MODIFIED, "Running", PodStatus.RUNNING),
+    Transition(PodStatus.INITIAL, PodEventType.MODIFIED, "Succeeded", PodStatus.SUCCEEDED),
+    Transition(PodStatus.INITIAL, PodEventType.MODIFIED, "Failed", PodStatus.FAILED, True),
+    Transition(PodStatus.INITIAL, PodEventType.ADDED, "Succeeded", PodStatus.SUCCEEDED),
+    Transition(PodStatus.INITIAL, PodEventType.ADDED, "Failed", PodStatus.FAILED, True),
+    Transition(PodStatus.INITIAL, PodEventType.DELETED, None, PodStatus.DELETED, True),
     Transition(PodStatus.PENDING, PodEventType.MODIFIED, "Running", PodStatus.RUNNING),
     Transition(PodStatus.PENDING, PodEventType.MODIFIED, "Succeeded", PodStatus.SUCCEEDED),
     Transition(PodStatus.PENDING, PodEventType.MODIFIED, "Failed", PodStatus.FAILED, True),
