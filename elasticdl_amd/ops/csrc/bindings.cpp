@@ -5,13 +5,9 @@
 
 #include <torch/extension.h>
 
-#include <ATen/cuda/CUDAContext.h>
+#include <c10/hip/HIPStream.h>
 
 #include <cstdint>
-
-// hipStream_t without pulling in hip_runtime.h
-struct ihipStream_t;
-typedef struct ihipStream_t* hipStream_t;
 
 extern "C" {
 void edl_dense_sgd(float*, const float*, int64_t, float, hipStream_t);
@@ -35,6 +31,25 @@ void edl_sparse_adagrad(float*, float*, const float*, const int32_t*, int64_t,
 void edl_sparse_ftrl(float*, float*, float*, const float*, const int32_t*,
                      int64_t, int64_t, float, float, float, float,
                      hipStream_t);
+void edl_dense_rmsprop(float*, float*, float*, float*, const float*, int64_t,
+                       float, float, float, float, hipStream_t);
+void edl_dense_adadelta(float*, float*, float*, const float*, int64_t, float,
+                        float, float, hipStream_t);
+void edl_dense_adamax(float*, float*, float*, const float*, int64_t, float,
+                      float, float, float, hipStream_t);
+void edl_dense_nadam(float*, float*, float*, const float*, int64_t, float,
+                     float, float, float, float, float, float, hipStream_t);
+void edl_sparse_rmsprop(float*, float*, float*, float*, const float*,
+                        const int32_t*, int64_t, int64_t, float, float, float,
+                        float, hipStream_t);
+void edl_sparse_adadelta(float*, float*, float*, const float*, const int32_t*,
+                         int64_t, int64_t, float, float, float, hipStream_t);
+void edl_sparse_adamax(float*, float*, float*, const float*, const int32_t*,
+                       int64_t, int64_t, float, float, float, float,
+                       hipStream_t);
+void edl_sparse_nadam(float*, float*, float*, const float*, const int32_t*,
+                      int64_t, int64_t, float, float, float, float, float,
+                      float, float, hipStream_t);
 void edl_ht_lookup_or_insert(int64_t*, int32_t*, int64_t, int32_t*, int32_t,
                              const int64_t*, int64_t, int32_t*, uint8_t*,
                              int32_t*, int64_t*, hipStream_t);
@@ -50,7 +65,7 @@ void edl_batch_compact(int32_t*, int32_t*, int64_t, int32_t*, const int32_t*,
 void edl_accumulate_rows(const float*, const int32_t*, int64_t, int64_t,
                          float*, hipStream_t);
 void edl_init_new_rows(float*, const int32_t*, const uint8_t*,
-                       const int64_t*, int64_t, int64_t, uint64_t, float,
+                       const int64_t*, int64_t, int64_t, uint64_t, int, float,
                        float, hipStream_t);  // is_new may be null: slot<0 skips
 void edl_gather_rows(const float*, const int32_t*, int64_t, int64_t, float*,
                      hipStream_t);
@@ -68,8 +83,9 @@ void edl_fused_adamw_bf16(void*, float*, float*, float*, const void*, int64_t,
 namespace {
 
 hipStream_t cur_stream() {
-  return reinterpret_cast<hipStream_t>(
-      at::cuda::getCurrentCUDAStream().stream());
+  // the explicit HIP accessor (torch's current stream for device 0's
+  // active context) — kernels launch on the same stream torch uses
+  return c10::hip::getCurrentHIPStream().stream();
 }
 
 void check_f32_cuda(const torch::Tensor& t, const char* name) {
@@ -119,6 +135,43 @@ void dense_ftrl(torch::Tensor p, torch::Tensor z, torch::Tensor n,
   edl_dense_ftrl(p.data_ptr<float>(), z.data_ptr<float>(), n.data_ptr<float>(),
                  g.data_ptr<float>(), p.numel(), alpha, beta, l1, l2,
                  cur_stream());
+}
+
+void dense_rmsprop(torch::Tensor p, torch::Tensor ms, torch::Tensor mom,
+                   c10::optional<torch::Tensor> mg, torch::Tensor g,
+                   double lr, double rho, double momentum, double eps) {
+  check_f32_cuda(p, "param");
+  edl_dense_rmsprop(p.data_ptr<float>(), ms.data_ptr<float>(),
+                    mom.data_ptr<float>(),
+                    mg.has_value() ? mg->data_ptr<float>() : nullptr,
+                    g.data_ptr<float>(), p.numel(), lr, rho, momentum, eps,
+                    cur_stream());
+}
+
+void dense_adadelta(torch::Tensor p, torch::Tensor ag, torch::Tensor au,
+                    torch::Tensor g, double lr, double rho, double eps) {
+  check_f32_cuda(p, "param");
+  edl_dense_adadelta(p.data_ptr<float>(), ag.data_ptr<float>(),
+                     au.data_ptr<float>(), g.data_ptr<float>(), p.numel(), lr,
+                     rho, eps, cur_stream());
+}
+
+void dense_adamax(torch::Tensor p, torch::Tensor m, torch::Tensor v,
+                  torch::Tensor g, double lr_t, double b1, double b2,
+                  double eps) {
+  check_f32_cuda(p, "param");
+  edl_dense_adamax(p.data_ptr<float>(), m.data_ptr<float>(),
+                   v.data_ptr<float>(), g.data_ptr<float>(), p.numel(), lr_t,
+                   b1, b2, eps, cur_stream());
+}
+
+void dense_nadam(torch::Tensor p, torch::Tensor m, torch::Tensor v,
+                 torch::Tensor g, double lr, double c1, double c2,
+                 double vcorr, double b1, double b2, double eps) {
+  check_f32_cuda(p, "param");
+  edl_dense_nadam(p.data_ptr<float>(), m.data_ptr<float>(),
+                  v.data_ptr<float>(), g.data_ptr<float>(), p.numel(), lr, c1,
+                  c2, vcorr, b1, b2, eps, cur_stream());
 }
 
 // ------------------------------ sparse optimizers -----------------------
@@ -176,6 +229,49 @@ void sparse_ftrl(torch::Tensor arena, torch::Tensor z, torch::Tensor n,
                   n.data_ptr<float>(), g.data_ptr<float>(),
                   slots.data_ptr<int32_t>(), g.size(0), g.size(1), alpha, beta,
                   l1, l2, cur_stream());
+}
+
+void sparse_rmsprop(torch::Tensor arena, torch::Tensor ms, torch::Tensor mom,
+                    c10::optional<torch::Tensor> mg, torch::Tensor g,
+                    torch::Tensor slots, double lr, double rho,
+                    double momentum, double eps) {
+  check_sparse(arena, g, slots);
+  edl_sparse_rmsprop(arena.data_ptr<float>(), ms.data_ptr<float>(),
+                     mom.data_ptr<float>(),
+                     mg.has_value() ? mg->data_ptr<float>() : nullptr,
+                     g.data_ptr<float>(), slots.data_ptr<int32_t>(),
+                     g.size(0), g.size(1), lr, rho, momentum, eps,
+                     cur_stream());
+}
+
+void sparse_adadelta(torch::Tensor arena, torch::Tensor ag, torch::Tensor au,
+                     torch::Tensor g, torch::Tensor slots, double lr,
+                     double rho, double eps) {
+  check_sparse(arena, g, slots);
+  edl_sparse_adadelta(arena.data_ptr<float>(), ag.data_ptr<float>(),
+                      au.data_ptr<float>(), g.data_ptr<float>(),
+                      slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr,
+                      rho, eps, cur_stream());
+}
+
+void sparse_adamax(torch::Tensor arena, torch::Tensor m, torch::Tensor v,
+                   torch::Tensor g, torch::Tensor slots, double lr_t,
+                   double b1, double b2, double eps) {
+  check_sparse(arena, g, slots);
+  edl_sparse_adamax(arena.data_ptr<float>(), m.data_ptr<float>(),
+                    v.data_ptr<float>(), g.data_ptr<float>(),
+                    slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr_t, b1,
+                    b2, eps, cur_stream());
+}
+
+void sparse_nadam(torch::Tensor arena, torch::Tensor m, torch::Tensor v,
+                  torch::Tensor g, torch::Tensor slots, double lr, double c1,
+                  double c2, double vcorr, double b1, double b2, double eps) {
+  check_sparse(arena, g, slots);
+  edl_sparse_nadam(arena.data_ptr<float>(), m.data_ptr<float>(),
+                   v.data_ptr<float>(), g.data_ptr<float>(),
+                   slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr, c1,
+                   c2, vcorr, b1, b2, eps, cur_stream());
 }
 
 // ------------------------------ hash table ------------------------------
@@ -251,13 +347,15 @@ void ht_lookup(torch::Tensor keys, torch::Tensor vals, torch::Tensor ids,
 
 void init_new_rows(torch::Tensor arena, torch::Tensor slots,
                    c10::optional<torch::Tensor> is_new, torch::Tensor ids,
-                   int64_t seed, double lo, double hi) {
+                   int64_t seed, int64_t mode, double a, double b) {
   check_f32_cuda(arena, "arena");
   TORCH_CHECK(ids.scalar_type() == torch::kInt64);
+  TORCH_CHECK(mode >= 0 && mode <= 3, "init mode must be 0..3");
   edl_init_new_rows(arena.data_ptr<float>(), slots.data_ptr<int32_t>(),
                     is_new.has_value() ? is_new->data_ptr<uint8_t>() : nullptr,
                     ids.data_ptr<int64_t>(), slots.numel(), arena.size(1),
-                    static_cast<uint64_t>(seed), lo, hi, cur_stream());
+                    static_cast<uint64_t>(seed), static_cast<int>(mode), a, b,
+                    cur_stream());
 }
 
 torch::Tensor gather_rows(torch::Tensor arena, torch::Tensor slots) {
@@ -342,6 +440,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sparse_adam", &sparse_adam);
   m.def("sparse_adagrad", &sparse_adagrad);
   m.def("sparse_ftrl", &sparse_ftrl);
+  m.def("dense_rmsprop", &dense_rmsprop);
+  m.def("dense_adadelta", &dense_adadelta);
+  m.def("dense_adamax", &dense_adamax);
+  m.def("dense_nadam", &dense_nadam);
+  m.def("sparse_rmsprop", &sparse_rmsprop);
+  m.def("sparse_adadelta", &sparse_adadelta);
+  m.def("sparse_adamax", &sparse_adamax);
+  m.def("sparse_nadam", &sparse_nadam);
   m.def("ht_lookup_or_insert", &ht_lookup_or_insert);
   m.def("ht_lookup", &ht_lookup);
   m.def("ht_insert_dup", &ht_insert_dup);

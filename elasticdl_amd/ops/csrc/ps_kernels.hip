@@ -188,6 +188,147 @@ struct FtrlOp {
   }
 };
 
+// TF ApplyRMSProp semantics (the reference wraps Keras RMSprop with slots
+// rms/momentum/mg, ps/optimizer_wrapper.py:139-145):
+//   ms = rho*ms + (1-rho)*g^2
+//   [centered] mg = rho*mg + (1-rho)*g ; denom = ms - mg^2
+//   mom = momentum*mom + lr*g/sqrt(denom + eps) ; p -= mom
+struct RmspropOp {
+  float* p;
+  float* ms;
+  float* mom;
+  float* mg;  // nullptr unless centered
+  float lr, rho, momentum, eps;
+  __device__ void one(float& pm, float g, float& s, float& mo, float& cg) {
+    s = rho * s + (1.f - rho) * g * g;
+    float denom = s;
+    if (mg != nullptr) {
+      cg = rho * cg + (1.f - rho) * g;
+      denom = s - cg * cg;
+    }
+    mo = momentum * mo + lr * g * rsqrtf(denom + eps);
+    pm -= mo;
+  }
+  __device__ void apply(int64_t off, float g) {
+    float pm = p[off], s = ms[off], mo = mom[off];
+    float cg = mg ? mg[off] : 0.f;
+    one(pm, g, s, mo, cg);
+    p[off] = pm; ms[off] = s; mom[off] = mo;
+    if (mg) mg[off] = cg;
+  }
+  __device__ void apply4(int64_t off, float4 g) {
+    float4* pp = reinterpret_cast<float4*>(p + off);
+    float4* sp = reinterpret_cast<float4*>(ms + off);
+    float4* op = reinterpret_cast<float4*>(mom + off);
+    float4 pr = *pp, s = *sp, mo = *op;
+    float4 cg = mg ? *reinterpret_cast<float4*>(mg + off)
+                   : make_float4(0.f, 0.f, 0.f, 0.f);
+    one(pr.x, g.x, s.x, mo.x, cg.x);
+    one(pr.y, g.y, s.y, mo.y, cg.y);
+    one(pr.z, g.z, s.z, mo.z, cg.z);
+    one(pr.w, g.w, s.w, mo.w, cg.w);
+    *pp = pr; *sp = s; *op = mo;
+    if (mg) *reinterpret_cast<float4*>(mg + off) = cg;
+  }
+};
+
+// TF ApplyAdadelta (slots accum_grad/accum_var,
+// optimizer_wrapper.py:122-126):
+//   ag = rho*ag + (1-rho)*g^2
+//   upd = sqrt(au + eps)/sqrt(ag + eps) * g
+//   au = rho*au + (1-rho)*upd^2 ; p -= lr*upd
+struct AdadeltaOp {
+  float* p;
+  float* ag;  // accum_grad
+  float* au;  // accum_var (accumulated update)
+  float lr, rho, eps;
+  __device__ void one(float& pm, float g, float& a, float& u) {
+    a = rho * a + (1.f - rho) * g * g;
+    float upd = sqrtf(u + eps) * rsqrtf(a + eps) * g;
+    u = rho * u + (1.f - rho) * upd * upd;
+    pm -= lr * upd;
+  }
+  __device__ void apply(int64_t off, float g) {
+    float pm = p[off], a = ag[off], u = au[off];
+    one(pm, g, a, u);
+    p[off] = pm; ag[off] = a; au[off] = u;
+  }
+  __device__ void apply4(int64_t off, float4 g) {
+    float4* pp = reinterpret_cast<float4*>(p + off);
+    float4* ap = reinterpret_cast<float4*>(ag + off);
+    float4* up = reinterpret_cast<float4*>(au + off);
+    float4 pr = *pp, a = *ap, u = *up;
+    one(pr.x, g.x, a.x, u.x);
+    one(pr.y, g.y, a.y, u.y);
+    one(pr.z, g.z, a.z, u.z);
+    one(pr.w, g.w, a.w, u.w);
+    *pp = pr; *ap = a; *up = u;
+  }
+};
+
+// Keras Adamax (slots m/v):
+//   m = b1*m + (1-b1)*g ; v = max(b2*v, |g|)
+//   p -= lr/(1-b1^t) * m / (v + eps)   [lr_t precomputed on host]
+struct AdamaxOp {
+  float* p;
+  float* m;
+  float* v;
+  float lr_t, b1, b2, eps;
+  __device__ void one(float& pm, float g, float& mm, float& vv) {
+    mm = b1 * mm + (1.f - b1) * g;
+    vv = fmaxf(b2 * vv, fabsf(g));
+    pm -= lr_t * mm / (vv + eps);
+  }
+  __device__ void apply(int64_t off, float g) {
+    float pm = p[off], mm = m[off], vv = v[off];
+    one(pm, g, mm, vv);
+    p[off] = pm; m[off] = mm; v[off] = vv;
+  }
+  __device__ void apply4(int64_t off, float4 g) {
+    float4* pp = reinterpret_cast<float4*>(p + off);
+    float4* mp = reinterpret_cast<float4*>(m + off);
+    float4* vp = reinterpret_cast<float4*>(v + off);
+    float4 pr = *pp, mm = *mp, vv = *vp;
+    one(pr.x, g.x, mm.x, vv.x);
+    one(pr.y, g.y, mm.y, vv.y);
+    one(pr.z, g.z, mm.z, vv.z);
+    one(pr.w, g.w, mm.w, vv.w);
+    *pp = pr; *mp = mm; *vp = vv;
+  }
+};
+
+// Nadam (Dozat 2016; slots m/v). Host precomputes
+//   c1 = b1/(1-b1^(t+1)),  c2 = (1-b1)/(1-b1^t),  vcorr = 1/(1-b2^t):
+//   m = b1*m + (1-b1)*g ; v = b2*v + (1-b2)*g^2
+//   p -= lr * (c1*m + c2*g) / (sqrt(v*vcorr) + eps)
+struct NadamOp {
+  float* p;
+  float* m;
+  float* v;
+  float lr, c1, c2, vcorr, b1, b2, eps;
+  __device__ void one(float& pm, float g, float& mm, float& vv) {
+    mm = b1 * mm + (1.f - b1) * g;
+    vv = b2 * vv + (1.f - b2) * g * g;
+    pm -= lr * (c1 * mm + c2 * g) / (sqrtf(vv * vcorr) + eps);
+  }
+  __device__ void apply(int64_t off, float g) {
+    float pm = p[off], mm = m[off], vv = v[off];
+    one(pm, g, mm, vv);
+    p[off] = pm; m[off] = mm; v[off] = vv;
+  }
+  __device__ void apply4(int64_t off, float4 g) {
+    float4* pp = reinterpret_cast<float4*>(p + off);
+    float4* mp = reinterpret_cast<float4*>(m + off);
+    float4* vp = reinterpret_cast<float4*>(v + off);
+    float4 pr = *pp, mm = *mp, vv = *vp;
+    one(pr.x, g.x, mm.x, vv.x);
+    one(pr.y, g.y, mm.y, vv.y);
+    one(pr.z, g.z, mm.z, vv.z);
+    one(pr.w, g.w, mm.w, vv.w);
+    *pp = pr; *mp = mm; *vp = vv;
+  }
+};
+
 // ---------------------------------------------------------------------------
 // Generic update kernels.
 //
@@ -299,6 +440,30 @@ void edl_dense_ftrl(float* p, float* z, float* n, const float* g,
   launch_dense(FtrlOp{p, z, n, alpha, beta, l1, l2}, g, numel, s);
 }
 
+void edl_dense_rmsprop(float* p, float* ms, float* mom, float* mg,
+                       const float* g, int64_t numel, float lr, float rho,
+                       float momentum, float eps, hipStream_t s) {
+  launch_dense(RmspropOp{p, ms, mom, mg, lr, rho, momentum, eps}, g, numel, s);
+}
+
+void edl_dense_adadelta(float* p, float* ag, float* au, const float* g,
+                        int64_t numel, float lr, float rho, float eps,
+                        hipStream_t s) {
+  launch_dense(AdadeltaOp{p, ag, au, lr, rho, eps}, g, numel, s);
+}
+
+void edl_dense_adamax(float* p, float* m, float* v, const float* g,
+                      int64_t numel, float lr_t, float b1, float b2, float eps,
+                      hipStream_t s) {
+  launch_dense(AdamaxOp{p, m, v, lr_t, b1, b2, eps}, g, numel, s);
+}
+
+void edl_dense_nadam(float* p, float* m, float* v, const float* g,
+                     int64_t numel, float lr, float c1, float c2, float vcorr,
+                     float b1, float b2, float eps, hipStream_t s) {
+  launch_dense(NadamOp{p, m, v, lr, c1, c2, vcorr, b1, b2, eps}, g, numel, s);
+}
+
 void edl_sparse_sgd(float* arena, const float* g, const int32_t* slots,
                     int64_t n, int64_t dim, float lr, hipStream_t s) {
   launch_sparse(SgdOp{arena, lr}, g, slots, n, dim, s);
@@ -329,6 +494,35 @@ void edl_sparse_ftrl(float* arena, float* z, float* nacc, const float* g,
                      float beta, float l1, float l2, hipStream_t s) {
   launch_sparse(FtrlOp{arena, z, nacc, alpha, beta, l1, l2}, g, slots, n, dim,
                 s);
+}
+
+void edl_sparse_rmsprop(float* arena, float* ms, float* mom, float* mg,
+                        const float* g, const int32_t* slots, int64_t n,
+                        int64_t dim, float lr, float rho, float momentum,
+                        float eps, hipStream_t s) {
+  launch_sparse(RmspropOp{arena, ms, mom, mg, lr, rho, momentum, eps}, g,
+                slots, n, dim, s);
+}
+
+void edl_sparse_adadelta(float* arena, float* ag, float* au, const float* g,
+                         const int32_t* slots, int64_t n, int64_t dim,
+                         float lr, float rho, float eps, hipStream_t s) {
+  launch_sparse(AdadeltaOp{arena, ag, au, lr, rho, eps}, g, slots, n, dim, s);
+}
+
+void edl_sparse_adamax(float* arena, float* m, float* v, const float* g,
+                       const int32_t* slots, int64_t n, int64_t dim,
+                       float lr_t, float b1, float b2, float eps,
+                       hipStream_t s) {
+  launch_sparse(AdamaxOp{arena, m, v, lr_t, b1, b2, eps}, g, slots, n, dim, s);
+}
+
+void edl_sparse_nadam(float* arena, float* m, float* v, const float* g,
+                      const int32_t* slots, int64_t n, int64_t dim, float lr,
+                      float c1, float c2, float vcorr, float b1, float b2,
+                      float eps, hipStream_t s) {
+  launch_sparse(NadamOp{arena, m, v, lr, c1, c2, vcorr, b1, b2, eps}, g, slots,
+                n, dim, s);
 }
 
 }  // extern "C"
@@ -569,17 +763,63 @@ __global__ void ht_lookup_kernel(const int64_t* __restrict__ keys,
   }
 }
 
-// Lazy row init: uniform RNG in [lo, hi) from a stateless splitmix hash of
-// (seed, ID, col) — keyed on the embedding ID, not the arena slot, so the
-// value of a row is deterministic regardless of the (atomics-ordered,
-// nondeterministic) slot assignment. Reproduces the reference's
-// init-on-first-touch (embedding_table.go:40-58) deterministically.
+// Lazy row init from a stateless splitmix hash of (seed, ID, col) — keyed
+// on the embedding ID, not the arena slot, so a row's value is
+// deterministic regardless of the (atomics-ordered, nondeterministic)
+// slot assignment. Reproduces the reference's init-on-first-touch
+// (embedding_table.go:40-58) with the full initializer set of
+// go/pkg/common/initializer.go:60-155:
+//   mode 0 uniform[a, b)        mode 1 normal(mean=a, std=b)
+//   mode 2 truncated_normal     mode 3 constant(a)
+// Truncated normal resamples (fresh sub-seeds) until |z| <= 2, max 16
+// tries then clamps — matching the TF definition the reference uses.
+#define EDL_INIT_UNIFORM 0
+#define EDL_INIT_NORMAL 1
+#define EDL_INIT_TRUNC_NORMAL 2
+#define EDL_INIT_CONSTANT 3
+
+__device__ inline float edl_u01(uint64_t r) {
+  return (float)(r >> 40) * (1.0f / 16777216.0f);  // [0,1) from top 24 bits
+}
+
+__device__ inline float edl_normal_z(uint64_t x, uint64_t col, int k) {
+  // Box-Muller; u1 in (0,1] so logf never sees 0
+  uint64_t r1 = edl_hash_u64(x ^ (col + (uint64_t)k * 0x632BE59Bull));
+  uint64_t r2 = edl_hash_u64(r1 ^ 0xDA3E0B5Cull);
+  float u1 = ((float)(r1 >> 40) + 1.0f) * (1.0f / 16777216.0f);
+  float u2 = edl_u01(r2);
+  return sqrtf(-2.0f * logf(u1)) * cosf(6.2831853071795864f * u2);
+}
+
+__device__ inline float edl_init_value(uint64_t x, uint64_t col, int mode,
+                                       float a, float b) {
+  switch (mode) {
+    case EDL_INIT_NORMAL:
+      return a + b * edl_normal_z(x, col, 0);
+    case EDL_INIT_TRUNC_NORMAL: {
+      float z = 0.f;
+      for (int k = 0; k < 16; ++k) {
+        z = edl_normal_z(x, col, k);
+        if (fabsf(z) <= 2.0f) break;
+      }
+      z = fminf(fmaxf(z, -2.0f), 2.0f);
+      return a + b * z;
+    }
+    case EDL_INIT_CONSTANT:
+      return a;
+    default: {  // uniform
+      uint64_t r = edl_hash_u64(x ^ col);
+      return a + edl_u01(r) * (b - a);
+    }
+  }
+}
+
 __global__ void init_new_rows_kernel(float* __restrict__ arena,
                                      const int32_t* __restrict__ slots,
                                      const uint8_t* __restrict__ is_new,
                                      const int64_t* __restrict__ ids,
                                      int64_t n, int64_t dim, uint64_t seed,
-                                     float lo, float hi) {
+                                     int mode, float a, float b) {
   int64_t total = n * dim;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -589,9 +829,7 @@ __global__ void init_new_rows_kernel(float* __restrict__ arena,
     int64_t col = i - row * dim;
     int64_t slot = slots[row];
     uint64_t x = edl_hash_u64(seed ^ (uint64_t)ids[row]);
-    uint64_t r = edl_hash_u64(x ^ (uint64_t)col);
-    float u = (float)(r >> 40) * (1.0f / 16777216.0f);  // [0,1) from top 24 bits
-    arena[slot * dim + col] = lo + u * (hi - lo);
+    arena[slot * dim + col] = edl_init_value(x, (uint64_t)col, mode, a, b);
   }
 }
 
@@ -698,10 +936,10 @@ void edl_ht_lookup(const int64_t* keys, const int32_t* vals, int64_t capacity,
 
 void edl_init_new_rows(float* arena, const int32_t* slots,
                        const uint8_t* is_new, const int64_t* ids, int64_t n,
-                       int64_t dim, uint64_t seed, float lo, float hi,
+                       int64_t dim, uint64_t seed, int mode, float a, float b,
                        hipStream_t s) {
   init_new_rows_kernel<<<grid_for(n * dim), THREADS, 0, s>>>(
-      arena, slots, is_new, ids, n, dim, seed, lo, hi);
+      arena, slots, is_new, ids, n, dim, seed, mode, a, b);
 }
 
 void edl_gather_rows(const float* arena, const int32_t* slots, int64_t n,

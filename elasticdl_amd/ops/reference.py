@@ -61,6 +61,49 @@ def dense_ftrl(p, z, n, g, alpha, beta, l1, l2):
     p.copy_(new_p)
 
 
+def dense_rmsprop(p, ms, mom, mg, g, lr, rho, momentum, eps):
+    """TF ApplyRMSProp; mg=None -> non-centered."""
+    ms.mul_(rho).addcmul_(g, g, value=1 - rho)
+    denom = ms
+    if mg is not None:
+        mg.mul_(rho).add_(g, alpha=1 - rho)
+        denom = ms - mg * mg
+    mom.mul_(momentum).add_(lr * g / (denom + eps).sqrt())
+    p.sub_(mom)
+
+
+def dense_adadelta(p, ag, au, g, lr, rho, eps):
+    ag.mul_(rho).addcmul_(g, g, value=1 - rho)
+    upd = (au + eps).sqrt() / (ag + eps).sqrt() * g
+    au.mul_(rho).addcmul_(upd, upd, value=1 - rho)
+    p.sub_(lr * upd)
+
+
+def adamax_lr_t(lr: float, step: int, b1: float) -> float:
+    return lr / (1.0 - b1 ** step)
+
+
+def dense_adamax(p, m, v, g, lr_t, b1, b2, eps):
+    m.mul_(b1).add_(g, alpha=1 - b1)
+    torch.maximum(b2 * v, g.abs(), out=v)
+    p.sub_(lr_t * m / (v + eps))
+
+
+def nadam_coeffs(step: int, b1: float, b2: float):
+    """(c1, c2, vcorr) for the Dozat-Nadam update (see NadamOp)."""
+    return (
+        b1 / (1.0 - b1 ** (step + 1)),
+        (1.0 - b1) / (1.0 - b1 ** step),
+        1.0 / (1.0 - b2 ** step),
+    )
+
+
+def dense_nadam(p, m, v, g, lr, c1, c2, vcorr, b1, b2, eps):
+    m.mul_(b1).add_(g, alpha=1 - b1)
+    v.mul_(b2).addcmul_(g, g, value=1 - b2)
+    p.sub_(lr * (c1 * m + c2 * g) / ((v * vcorr).sqrt() + eps))
+
+
 # ------------------------------- sparse ----------------------------------
 def _rows(arena, slots):
     return arena.index_select(0, slots.long())
@@ -115,6 +158,52 @@ def sparse_ftrl(arena, z, n, g, slots, alpha, beta, l1, l2):
     arena.index_copy_(0, s, new_p)
 
 
+def sparse_rmsprop(arena, ms, mom, mg, g, slots, lr, rho, momentum, eps):
+    s = slots.long()
+    sv = rho * _rows(ms, slots) + (1 - rho) * g * g
+    ms.index_copy_(0, s, sv)
+    denom = sv
+    if mg is not None:
+        cg = rho * _rows(mg, slots) + (1 - rho) * g
+        mg.index_copy_(0, s, cg)
+        denom = sv - cg * cg
+    mo = momentum * _rows(mom, slots) + lr * g / (denom + eps).sqrt()
+    mom.index_copy_(0, s, mo)
+    arena.index_copy_(0, s, _rows(arena, slots) - mo)
+
+
+def sparse_adadelta(arena, ag, au, g, slots, lr, rho, eps):
+    s = slots.long()
+    a = rho * _rows(ag, slots) + (1 - rho) * g * g
+    ag.index_copy_(0, s, a)
+    u0 = _rows(au, slots)
+    upd = (u0 + eps).sqrt() / (a + eps).sqrt() * g
+    au.index_copy_(0, s, rho * u0 + (1 - rho) * upd * upd)
+    arena.index_copy_(0, s, _rows(arena, slots) - lr * upd)
+
+
+def sparse_adamax(arena, m, v, g, slots, lr_t, b1, b2, eps):
+    s = slots.long()
+    mm = b1 * _rows(m, slots) + (1 - b1) * g
+    vv = torch.maximum(b2 * _rows(v, slots), g.abs())
+    m.index_copy_(0, s, mm)
+    v.index_copy_(0, s, vv)
+    arena.index_copy_(0, s, _rows(arena, slots) - lr_t * mm / (vv + eps))
+
+
+def sparse_nadam(arena, m, v, g, slots, lr, c1, c2, vcorr, b1, b2, eps):
+    s = slots.long()
+    mm = b1 * _rows(m, slots) + (1 - b1) * g
+    vv = b2 * _rows(v, slots) + (1 - b2) * g * g
+    m.index_copy_(0, s, mm)
+    v.index_copy_(0, s, vv)
+    arena.index_copy_(
+        0, s,
+        _rows(arena, slots)
+        - lr * (c1 * mm + c2 * g) / ((vv * vcorr).sqrt() + eps),
+    )
+
+
 # ---------------------------- rows / rng init -----------------------------
 def _splitmix64(x: torch.Tensor) -> torch.Tensor:
     """Vectorized splitmix64 finalizer matching edl_hash_u64 in
@@ -126,25 +215,63 @@ def _splitmix64(x: torch.Tensor) -> torch.Tensor:
     return x ^ (x >> 31)
 
 
-def init_rows_values(ids: torch.Tensor, dim: int, seed: int, lo: float, hi: float):
-    """Deterministic uniform rows, same bits as init_new_rows_kernel:
-    r = splitmix64(splitmix64(seed ^ id) ^ col), keyed on the embedding ID
-    so values are independent of arena slot assignment order."""
+# init modes (must match EDL_INIT_* in ps_kernels.hip)
+INIT_UNIFORM, INIT_NORMAL, INIT_TRUNC_NORMAL, INIT_CONSTANT = 0, 1, 2, 3
+
+
+def _sm64(v):
     import numpy as np
 
-    def sm64(v):
-        v = v + np.uint64(0x9E3779B97F4A7C15)
-        v = (v ^ (v >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
-        v = (v ^ (v >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
-        return v ^ (v >> np.uint64(31))
+    v = v + np.uint64(0x9E3779B97F4A7C15)
+    v = (v ^ (v >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+    v = (v ^ (v >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+    return v ^ (v >> np.uint64(31))
+
+
+def init_rows_values(ids: torch.Tensor, dim: int, seed: int,
+                     mode: int = INIT_UNIFORM, a: float = -0.05,
+                     b: float = 0.05):
+    """Deterministic rows, same generator as init_new_rows_kernel —
+    stateless splitmix64 keyed on (seed, embedding ID, col), independent of
+    arena slot order. Uniform is bit-identical to the GPU kernel; the
+    normal paths use the same integer draws (float libm differences are
+    within test tolerance). Modes mirror the reference initializer set
+    (go/pkg/common/initializer.go:60-155)."""
+    import numpy as np
 
     idv = ids.detach().cpu().to(torch.int64).view(-1, 1).numpy().astype(np.uint64)
     cols = np.arange(dim, dtype=np.uint64).reshape(1, -1)
     with np.errstate(over="ignore"):
-        x = sm64(np.uint64(seed) ^ idv)
-        r = sm64(x ^ cols)
-    u = (r >> np.uint64(40)).astype(np.float32) * np.float32(1.0 / 16777216.0)
-    vals = lo + u * (hi - lo)
+        x = _sm64(np.uint64(seed) ^ idv)
+        if mode == INIT_CONSTANT:
+            vals = np.full((idv.shape[0], dim), a, dtype=np.float32)
+        elif mode in (INIT_NORMAL, INIT_TRUNC_NORMAL):
+
+            def z_for(k):
+                r1 = _sm64(x ^ (cols + np.uint64(k * 0x632BE59B)))
+                r2 = _sm64(r1 ^ np.uint64(0xDA3E0B5C))
+                u1 = ((r1 >> np.uint64(40)).astype(np.float32) + 1.0) \
+                    * np.float32(1.0 / 16777216.0)
+                u2 = (r2 >> np.uint64(40)).astype(np.float32) \
+                    * np.float32(1.0 / 16777216.0)
+                return np.sqrt(-2.0 * np.log(u1)) * np.cos(
+                    np.float32(6.2831853071795864) * u2
+                )
+
+            z = z_for(0)
+            if mode == INIT_TRUNC_NORMAL:
+                for k in range(1, 16):
+                    bad = np.abs(z) > 2.0
+                    if not bad.any():
+                        break
+                    z = np.where(bad, z_for(k), z)
+                z = np.clip(z, -2.0, 2.0)
+            vals = (a + b * z).astype(np.float32)
+        else:  # uniform
+            r = _sm64(x ^ cols)
+            u = (r >> np.uint64(40)).astype(np.float32) \
+                * np.float32(1.0 / 16777216.0)
+            vals = (a + u * (b - a)).astype(np.float32)
     return torch.from_numpy(vals).to(ids.device)
 
 

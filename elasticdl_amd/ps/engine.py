@@ -106,7 +106,9 @@ class PSEngine:
     def _create_table(self, info: dict) -> None:
         import hashlib
 
-        init = info.get("initializer", ["uniform", -0.05, 0.05])
+        init = list(info.get("initializer", ["uniform", -0.05, 0.05]))
+        while len(init) < 3:
+            init.append(0.0)
         # per-table seed must be deterministic across processes/restarts
         # (Python's hash() is randomized per process via PYTHONHASHSEED,
         # which would break the id-keyed-init reproducibility guarantee)

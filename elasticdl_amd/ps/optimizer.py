@@ -82,6 +82,34 @@ class Optimizer:
                 float(args.get("l1", args.get("l1_regularization_strength", 0.0))),
                 float(args.get("l2", args.get("l2_regularization_strength", 0.0))),
             )
+        if t == "rmsprop":
+            return RmspropOptimizer(
+                lr,
+                float(args.get("rho", args.get("decay", 0.9))),
+                float(args.get("momentum", 0.0)),
+                float(args.get("epsilon", 1e-7)),
+                args.get("centered", "false").lower() in _TRUE,
+            )
+        if t == "adadelta":
+            return AdadeltaOptimizer(
+                lr,
+                float(args.get("rho", 0.95)),
+                float(args.get("epsilon", 1e-7)),
+            )
+        if t == "adamax":
+            return AdamaxOptimizer(
+                lr,
+                float(args.get("beta_1", 0.9)),
+                float(args.get("beta_2", 0.999)),
+                float(args.get("epsilon", 1e-7)),
+            )
+        if t == "nadam":
+            return NadamOptimizer(
+                lr,
+                float(args.get("beta_1", 0.9)),
+                float(args.get("beta_2", 0.999)),
+                float(args.get("epsilon", 1e-7)),
+            )
         raise ValueError(f"unknown optimizer type: {opt_type}")
 
     # -- common plumbing --------------------------------------------------
@@ -255,6 +283,162 @@ class AdagradOptimizer(Optimizer):
             _C.sparse_adagrad(table.arena, m, grads, slots, lr, self.eps)
         else:
             reference.sparse_adagrad(table.arena, m, grads, slots, lr, self.eps)
+
+
+class RmspropOptimizer(Optimizer):
+    """TF ApplyRMSProp semantics; slots rms/momentum(+mg when centered)
+    match the reference wrapper (ps/optimizer_wrapper.py:139-145)."""
+
+    def __init__(self, lr, rho, momentum, eps, centered):
+        super().__init__(lr)
+        self.rho, self.momentum, self.eps = rho, momentum, eps
+        self.centered = centered
+
+    @property
+    def SLOT_NAMES(self):
+        return ("rms", "momentum", "mg") if self.centered \
+            else ("rms", "momentum")
+
+    def apply_dense(self, name, param, grad, lr_mult=1.0):
+        st = self._dense_states(name, param)
+        mg = st.get("mg")
+        lr = self.base_lr * lr_mult
+        if self._native(param):
+            from elasticdl_amd.ops import _C
+
+            _C.dense_rmsprop(param, st["rms"], st["momentum"], mg,
+                             grad.contiguous(), lr, self.rho, self.momentum,
+                             self.eps)
+        else:
+            reference.dense_rmsprop(param, st["rms"], st["momentum"], mg,
+                                    grad, lr, self.rho, self.momentum,
+                                    self.eps)
+
+    def _apply_rows(self, table, grads, slots, lr_mult):
+        ms = table.get_slot_arena("rms")
+        mom = table.get_slot_arena("momentum")
+        mg = table.get_slot_arena("mg") if self.centered else None
+        lr = self.base_lr * lr_mult
+        if self._native(table.arena):
+            from elasticdl_amd.ops import _C
+
+            _C.sparse_rmsprop(table.arena, ms, mom, mg, grads, slots, lr,
+                              self.rho, self.momentum, self.eps)
+        else:
+            reference.sparse_rmsprop(table.arena, ms, mom, mg, grads, slots,
+                                     lr, self.rho, self.momentum, self.eps)
+
+
+class AdadeltaOptimizer(Optimizer):
+    SLOT_NAMES = ("accum_grad", "accum_var")
+
+    def __init__(self, lr, rho, eps):
+        super().__init__(lr)
+        self.rho, self.eps = rho, eps
+
+    def apply_dense(self, name, param, grad, lr_mult=1.0):
+        st = self._dense_states(name, param)
+        lr = self.base_lr * lr_mult
+        if self._native(param):
+            from elasticdl_amd.ops import _C
+
+            _C.dense_adadelta(param, st["accum_grad"], st["accum_var"],
+                              grad.contiguous(), lr, self.rho, self.eps)
+        else:
+            reference.dense_adadelta(param, st["accum_grad"],
+                                     st["accum_var"], grad, lr, self.rho,
+                                     self.eps)
+
+    def _apply_rows(self, table, grads, slots, lr_mult):
+        ag = table.get_slot_arena("accum_grad")
+        au = table.get_slot_arena("accum_var")
+        lr = self.base_lr * lr_mult
+        if self._native(table.arena):
+            from elasticdl_amd.ops import _C
+
+            _C.sparse_adadelta(table.arena, ag, au, grads, slots, lr,
+                               self.rho, self.eps)
+        else:
+            reference.sparse_adadelta(table.arena, ag, au, grads, slots, lr,
+                                      self.rho, self.eps)
+
+
+class AdamaxOptimizer(Optimizer):
+    SLOT_NAMES = ("m", "v")
+
+    def __init__(self, lr, b1, b2, eps):
+        super().__init__(lr)
+        self.b1, self.b2, self.eps = b1, b2, eps
+
+    def _lr_t(self, lr_mult):
+        return reference.adamax_lr_t(
+            self.base_lr * lr_mult, max(self.step, 1), self.b1
+        )
+
+    def apply_dense(self, name, param, grad, lr_mult=1.0):
+        st = self._dense_states(name, param)
+        lr_t = self._lr_t(lr_mult)
+        if self._native(param):
+            from elasticdl_amd.ops import _C
+
+            _C.dense_adamax(param, st["m"], st["v"], grad.contiguous(), lr_t,
+                            self.b1, self.b2, self.eps)
+        else:
+            reference.dense_adamax(param, st["m"], st["v"], grad, lr_t,
+                                   self.b1, self.b2, self.eps)
+
+    def _apply_rows(self, table, grads, slots, lr_mult):
+        m = table.get_slot_arena("m")
+        v = table.get_slot_arena("v")
+        lr_t = self._lr_t(lr_mult)
+        if self._native(table.arena):
+            from elasticdl_amd.ops import _C
+
+            _C.sparse_adamax(table.arena, m, v, grads, slots, lr_t, self.b1,
+                             self.b2, self.eps)
+        else:
+            reference.sparse_adamax(table.arena, m, v, grads, slots, lr_t,
+                                    self.b1, self.b2, self.eps)
+
+
+class NadamOptimizer(Optimizer):
+    """Dozat-Nadam (see NadamOp in ps_kernels.hip for the exact update)."""
+
+    SLOT_NAMES = ("m", "v")
+
+    def __init__(self, lr, b1, b2, eps):
+        super().__init__(lr)
+        self.b1, self.b2, self.eps = b1, b2, eps
+
+    def _coeffs(self):
+        return reference.nadam_coeffs(max(self.step, 1), self.b1, self.b2)
+
+    def apply_dense(self, name, param, grad, lr_mult=1.0):
+        st = self._dense_states(name, param)
+        c1, c2, vcorr = self._coeffs()
+        lr = self.base_lr * lr_mult
+        if self._native(param):
+            from elasticdl_amd.ops import _C
+
+            _C.dense_nadam(param, st["m"], st["v"], grad.contiguous(), lr,
+                           c1, c2, vcorr, self.b1, self.b2, self.eps)
+        else:
+            reference.dense_nadam(param, st["m"], st["v"], grad, lr, c1, c2,
+                                  vcorr, self.b1, self.b2, self.eps)
+
+    def _apply_rows(self, table, grads, slots, lr_mult):
+        m = table.get_slot_arena("m")
+        v = table.get_slot_arena("v")
+        c1, c2, vcorr = self._coeffs()
+        lr = self.base_lr * lr_mult
+        if self._native(table.arena):
+            from elasticdl_amd.ops import _C
+
+            _C.sparse_nadam(table.arena, m, v, grads, slots, lr, c1, c2,
+                            vcorr, self.b1, self.b2, self.eps)
+        else:
+            reference.sparse_nadam(table.arena, m, v, grads, slots, lr, c1,
+                                   c2, vcorr, self.b1, self.b2, self.eps)
 
 
 class FtrlOptimizer(Optimizer):

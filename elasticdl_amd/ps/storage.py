@@ -134,9 +134,9 @@ class EmbeddingTable:
                 self._error,
                 self._ids_by_slot,  # id bookkeeping done in-kernel
             )
-            lo, hi = self._init_range()
+            mode, a, b = self._init_params()
             self._C.init_new_rows(
-                self.arena, slots, is_new, unique_ids, self.seed, lo, hi
+                self.arena, slots, is_new, unique_ids, self.seed, mode, a, b
             )
             return slots
         # ----- CPU path
@@ -160,17 +160,35 @@ class EmbeddingTable:
         if new_slots:
             ns = torch.tensor([s for s, _ in new_slots], dtype=torch.int32)
             nids = torch.tensor([v for _, v in new_slots], dtype=torch.int64)
-            lo, hi = self._init_range()
+            mode, a, b = self._init_params()
             self.arena[ns.long()] = reference.init_rows_values(
-                nids, self.dim, self.seed, lo, hi
+                nids, self.dim, self.seed, mode, a, b
             )
         return slots
 
-    def _init_range(self) -> Tuple[float, float]:
+    _INIT_MODES = {
+        "uniform": reference.INIT_UNIFORM,
+        "random_uniform": reference.INIT_UNIFORM,
+        "normal": reference.INIT_NORMAL,
+        "random_normal": reference.INIT_NORMAL,
+        "truncated_normal": reference.INIT_TRUNC_NORMAL,
+        "zero": reference.INIT_CONSTANT,
+        "zeros": reference.INIT_CONSTANT,
+        "constant": reference.INIT_CONSTANT,
+    }
+
+    def _init_params(self) -> Tuple[int, float, float]:
+        """(mode, a, b) for the init kernel. Mirrors the reference's
+        per-table initializer set (go/pkg/common/initializer.go:60-155):
+        uniform(lo=a, hi=b), normal(mean=a, std=b), truncated_normal,
+        constant(a); slot tables are constant-initialized."""
         if self.is_slot:
-            return (self.slot_init_value, self.slot_init_value)
-        kind, lo, hi = self.initializer
-        return (lo, hi)
+            return reference.INIT_CONSTANT, float(self.slot_init_value), 0.0
+        kind, a, b = self.initializer
+        mode = self._INIT_MODES.get(str(kind).lower(), reference.INIT_UNIFORM)
+        if str(kind).lower() in ("zero", "zeros"):
+            a = 0.0
+        return mode, float(a), float(b)
 
     def lookup_or_create_dup(self, ids: torch.Tensor) -> torch.Tensor:
         """Duplicate-tolerant lookup/create: two kernel launches (insert
@@ -186,10 +204,10 @@ class EmbeddingTable:
         )
         slots = torch.empty(n, dtype=torch.int32, device=self.device)
         self._C.ht_lookup(self._keys, self._vals, ids, slots)
-        lo, hi = self._init_range()
+        mode, a, b = self._init_params()
         # is_new=None: rows with new_slots < 0 are skipped in-kernel
         self._C.init_new_rows(
-            self.arena, new_slots, None, ids, self.seed, lo, hi
+            self.arena, new_slots, None, ids, self.seed, mode, a, b
         )
         return slots
 

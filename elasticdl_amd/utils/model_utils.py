@@ -103,4 +103,34 @@ def get_optimizer_info(opt) -> tuple:
     if isinstance(opt, torch.optim.Adagrad):
         g = opt.param_groups[0]
         return ("adagrad", f"learning_rate={g['lr']};epsilon={g['eps']}")
+    if isinstance(opt, torch.optim.RMSprop):
+        g = opt.param_groups[0]
+        return (
+            "rmsprop",
+            f"learning_rate={g['lr']};rho={g['alpha']}"
+            f";momentum={g.get('momentum', 0)};epsilon={g['eps']}"
+            f";centered={str(bool(g.get('centered'))).lower()}",
+        )
+    if isinstance(opt, torch.optim.Adadelta):
+        g = opt.param_groups[0]
+        return (
+            "adadelta",
+            f"learning_rate={g['lr']};rho={g['rho']};epsilon={g['eps']}",
+        )
+    if isinstance(opt, torch.optim.NAdam):
+        g = opt.param_groups[0]
+        b1, b2 = g["betas"]
+        return (
+            "nadam",
+            f"learning_rate={g['lr']};beta_1={b1};beta_2={b2}"
+            f";epsilon={g['eps']}",
+        )
+    if isinstance(opt, torch.optim.Adamax):
+        g = opt.param_groups[0]
+        b1, b2 = g["betas"]
+        return (
+            "adamax",
+            f"learning_rate={g['lr']};beta_1={b1};beta_2={b2}"
+            f";epsilon={g['eps']}",
+        )
     raise ValueError(f"cannot map optimizer {type(opt)} to PS opt_type")

@@ -265,9 +265,9 @@ def test_init_rows_bit_identical_to_cpu(C):
     slots = torch.arange(10, dtype=torch.int32, device="cuda")
     ids = (torch.arange(10, dtype=torch.int64) * 977 + 13).cuda()
     is_new = torch.ones(10, dtype=torch.uint8, device="cuda")
-    C.init_new_rows(arena, slots, is_new, ids, seed, -0.05, 0.05)
+    C.init_new_rows(arena, slots, is_new, ids, seed, 0, -0.05, 0.05)
     torch.cuda.synchronize()
-    ref = reference.init_rows_values(ids.cpu(), dim, seed, -0.05, 0.05)
+    ref = reference.init_rows_values(ids.cpu(), dim, seed, 0, -0.05, 0.05)
     assert torch.allclose(arena[:10].cpu(), ref, atol=1e-8), (
         arena[:10].cpu() - ref
     ).abs().max()
@@ -335,3 +335,141 @@ def test_engine_gpu_matches_cpu_end_to_end():
     r_gpu = run("cuda")
     for c, g in zip(r_cpu, r_gpu):
         assert torch.allclose(c, g, atol=1e-6), (c - g).abs().max()
+
+
+# ------------------- round-2 optimizer breadth ---------------
+@pytest.mark.parametrize("centered", [False, True])
+def test_dense_rmsprop(C, centered):
+    n = 4099
+    p, ms, mom, g = rand(n), torch.zeros(n), torch.zeros(n), rand(n)
+    mg = torch.zeros(n) if centered else None
+    pg, msg, momg = p.cuda(), ms.cuda(), mom.cuda()
+    mgg = mg.cuda() if centered else None
+    for _ in range(3):
+        C.dense_rmsprop(pg, msg, momg, mgg, g.cuda(), 0.01, 0.9, 0.5, 1e-7)
+        reference.dense_rmsprop(p, ms, mom, mg, g, 0.01, 0.9, 0.5, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-5)
+    assert torch.allclose(momg.cpu(), mom, atol=1e-6)
+
+
+def test_dense_adadelta(C):
+    n = 5001
+    p, ag, au, g = rand(n), torch.zeros(n), torch.zeros(n), rand(n)
+    pg, agg, aug = p.cuda(), ag.cuda(), au.cuda()
+    for _ in range(3):
+        C.dense_adadelta(pg, agg, aug, g.cuda(), 1.0, 0.95, 1e-7)
+        reference.dense_adadelta(p, ag, au, g, 1.0, 0.95, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-6)
+    assert torch.allclose(aug.cpu(), au, atol=1e-7)
+
+
+def test_dense_adamax(C):
+    n = 4097
+    p, m, v, g = rand(n), torch.zeros(n), torch.zeros(n), rand(n)
+    pg, mg_, vg = p.cuda(), m.cuda(), v.cuda()
+    for step in range(1, 4):
+        lr_t = reference.adamax_lr_t(0.01, step, 0.9)
+        C.dense_adamax(pg, mg_, vg, g.cuda(), lr_t, 0.9, 0.999, 1e-7)
+        reference.dense_adamax(p, m, v, g, lr_t, 0.9, 0.999, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-5)
+    assert torch.allclose(vg.cpu(), v, atol=1e-6)
+
+
+def test_dense_nadam(C):
+    n = 4097
+    p, m, v, g = rand(n), torch.zeros(n), torch.zeros(n), rand(n)
+    pg, mg_, vg = p.cuda(), m.cuda(), v.cuda()
+    for step in range(1, 4):
+        c1, c2, vcorr = reference.nadam_coeffs(step, 0.9, 0.999)
+        C.dense_nadam(pg, mg_, vg, g.cuda(), 0.01, c1, c2, vcorr,
+                      0.9, 0.999, 1e-7)
+        reference.dense_nadam(p, m, v, g, 0.01, c1, c2, vcorr,
+                              0.9, 0.999, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(pg.cpu(), p, atol=1e-5)
+
+
+@pytest.mark.parametrize("dim", [8, 17])
+def test_sparse_rmsprop_adadelta_adamax_nadam(C, dim):
+    n = 20
+    slots = torch.randperm(100)[:n].to(torch.int32)
+    g = rand(n, dim)
+
+    arena, ms, mom = rand(100, dim), torch.zeros(100, dim), torch.zeros(100, dim)
+    ag_, msg, momg = arena.cuda(), ms.cuda(), mom.cuda()
+    C.sparse_rmsprop(ag_, msg, momg, None, g.cuda(), slots.cuda(),
+                     0.01, 0.9, 0.0, 1e-7)
+    reference.sparse_rmsprop(arena, ms, mom, None, g, slots, 0.01, 0.9,
+                             0.0, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(ag_.cpu(), arena, atol=1e-6)
+
+    arena, a1, a2 = rand(100, dim), torch.zeros(100, dim), torch.zeros(100, dim)
+    ag_, a1g, a2g = arena.cuda(), a1.cuda(), a2.cuda()
+    C.sparse_adadelta(ag_, a1g, a2g, g.cuda(), slots.cuda(), 1.0, 0.95, 1e-7)
+    reference.sparse_adadelta(arena, a1, a2, g, slots, 1.0, 0.95, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(ag_.cpu(), arena, atol=1e-6)
+
+    arena, m, v = rand(100, dim), torch.zeros(100, dim), torch.zeros(100, dim)
+    ag_, mg_, vg = arena.cuda(), m.cuda(), v.cuda()
+    lr_t = reference.adamax_lr_t(0.01, 1, 0.9)
+    C.sparse_adamax(ag_, mg_, vg, g.cuda(), slots.cuda(), lr_t, 0.9,
+                    0.999, 1e-7)
+    reference.sparse_adamax(arena, m, v, g, slots, lr_t, 0.9, 0.999, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(ag_.cpu(), arena, atol=1e-6)
+
+    arena, m, v = rand(100, dim), torch.zeros(100, dim), torch.zeros(100, dim)
+    ag_, mg_, vg = arena.cuda(), m.cuda(), v.cuda()
+    c1, c2, vcorr = reference.nadam_coeffs(1, 0.9, 0.999)
+    C.sparse_nadam(ag_, mg_, vg, g.cuda(), slots.cuda(), 0.01, c1, c2, vcorr,
+                   0.9, 0.999, 1e-7)
+    reference.sparse_nadam(arena, m, v, g, slots, 0.01, c1, c2, vcorr,
+                           0.9, 0.999, 1e-7)
+    torch.cuda.synchronize()
+    assert torch.allclose(ag_.cpu(), arena, atol=1e-6)
+
+
+# ------------------- round-2 initializer breadth -------------
+@pytest.mark.parametrize("mode,a,b", [
+    (reference.INIT_NORMAL, 0.0, 1.0),
+    (reference.INIT_TRUNC_NORMAL, 0.5, 0.2),
+    (reference.INIT_CONSTANT, 0.37, 0.0),
+])
+def test_init_modes_match_cpu_oracle(C, mode, a, b):
+    dim, seed, n = 16, 777, 64
+    arena = torch.zeros(n, dim, device="cuda")
+    slots = torch.arange(n, dtype=torch.int32, device="cuda")
+    ids = (torch.arange(n, dtype=torch.int64) * 7919 + 3).cuda()
+    is_new = torch.ones(n, dtype=torch.uint8, device="cuda")
+    C.init_new_rows(arena, slots, is_new, ids, seed, mode, a, b)
+    torch.cuda.synchronize()
+    ref = reference.init_rows_values(ids.cpu(), dim, seed, mode, a, b)
+    assert torch.allclose(arena.cpu(), ref, atol=1e-5), (
+        (arena.cpu() - ref).abs().max()
+    )
+
+
+def test_init_normal_statistics(C):
+    dim, n = 64, 4096
+    arena = torch.zeros(n, dim, device="cuda")
+    slots = torch.arange(n, dtype=torch.int32, device="cuda")
+    ids = torch.arange(n, dtype=torch.int64).cuda()
+    is_new = torch.ones(n, dtype=torch.uint8, device="cuda")
+    C.init_new_rows(arena, slots, is_new, ids, 1, reference.INIT_NORMAL,
+                    0.0, 1.0)
+    torch.cuda.synchronize()
+    vals = arena.cpu().reshape(-1)
+    assert abs(vals.mean().item()) < 0.01
+    assert abs(vals.std().item() - 1.0) < 0.01
+    # truncated: everything within 2 sigma, std < 1
+    C.init_new_rows(arena, slots, is_new, ids, 1,
+                    reference.INIT_TRUNC_NORMAL, 0.0, 1.0)
+    torch.cuda.synchronize()
+    vals = arena.cpu().reshape(-1)
+    assert vals.abs().max().item() <= 2.0 + 1e-6
+    assert vals.std().item() < 0.95

@@ -89,6 +89,11 @@ def test_ftrl_update_nonzero():
         ("adam", "learning_rate=0.01"),
         ("adagrad", "learning_rate=0.1"),
         ("ftrl", "learning_rate=0.5;beta=1.0;l1=0.001;l2=0.001"),
+        ("rmsprop", "learning_rate=0.01;rho=0.9;momentum=0.5"),
+        ("rmsprop", "learning_rate=0.01;rho=0.9;centered=true"),
+        ("adadelta", "learning_rate=1.0;rho=0.95"),
+        ("adamax", "learning_rate=0.01"),
+        ("nadam", "learning_rate=0.01"),
     ],
 )
 def test_sparse_matches_dense_math(opt_type, opt_args):
@@ -135,3 +140,86 @@ def test_sparse_dedup_sums_before_apply():
 
     # summed grad = 2 -> m=4, p -= 2/sqrt(4) = 1
     assert torch.allclose(table.gather(torch.tensor([5])), before - 1.0)
+
+
+def test_rmsprop_matches_tf_semantics():
+    """Non-centered, no momentum: p -= lr*g/sqrt(rho*0+(1-rho)*g^2 + eps)."""
+    opt = Optimizer.create("rmsprop", "learning_rate=0.1;rho=0.9;epsilon=0.0")
+    p = torch.zeros(2)
+    g = torch.tensor([3.0, -4.0])
+    opt.begin_apply()
+    opt.apply_dense("p", p, g)
+    # ms = 0.1*g^2 ; upd = 0.1*g/sqrt(0.1*g^2) = 0.1*sign(g)*sqrt(10)
+    expect = -0.1 * torch.sign(g) * math.sqrt(10.0)
+    assert torch.allclose(p, expect, atol=1e-5)
+
+
+def test_adadelta_first_step():
+    opt = Optimizer.create("adadelta", "learning_rate=1.0;rho=0.9;epsilon=1e-6")
+    p = torch.zeros(3)
+    g = torch.ones(3)
+    opt.begin_apply()
+    opt.apply_dense("p", p, g)
+    # ag = 0.1 ; upd = sqrt(1e-6)/sqrt(0.1+1e-6)*1
+    expect = -math.sqrt(1e-6) / math.sqrt(0.1 + 1e-6)
+    assert torch.allclose(p, torch.full((3,), expect), atol=1e-7)
+
+
+def test_adamax_infinity_norm():
+    opt = Optimizer.create("adamax", "learning_rate=0.01;epsilon=0.0")
+    p = torch.zeros(2)
+    opt.begin_apply()
+    opt.apply_dense("p", p, torch.tensor([1.0, -2.0]))
+    # m = 0.1*g ; v = max(0, |g|) = |g| ; p -= lr/(1-0.9) * m/v
+    # = 0.1 * 0.1*g/|g| = 0.01*sign(g)
+    assert torch.allclose(p, torch.tensor([-0.01, 0.01]), atol=1e-6)
+
+
+def test_nadam_converges_on_quadratic():
+    opt = Optimizer.create("nadam", "learning_rate=0.1")
+    p = torch.tensor([5.0])
+    for _ in range(200):
+        opt.begin_apply()
+        opt.apply_dense("p", p, 2 * p.clone())  # grad of p^2
+    assert p.abs().item() < 0.05
+
+
+def test_factory_new_optimizers_and_slots():
+    assert Optimizer.create("rmsprop", "centered=true").SLOT_NAMES == (
+        "rms", "momentum", "mg")
+    assert Optimizer.create("rmsprop", "").SLOT_NAMES == ("rms", "momentum")
+    assert Optimizer.create("adadelta", "").SLOT_NAMES == (
+        "accum_grad", "accum_var")
+    assert Optimizer.create("adamax", "").SLOT_NAMES == ("m", "v")
+    assert Optimizer.create("nadam", "").SLOT_NAMES == ("m", "v")
+
+
+def test_initializer_modes_cpu():
+    from elasticdl_amd.ops import reference
+
+    ids = torch.tensor([1, 2, 3], dtype=torch.int64)
+    const = reference.init_rows_values(ids, 4, 0, reference.INIT_CONSTANT,
+                                       0.25, 0.0)
+    assert torch.all(const == 0.25)
+    norm = reference.init_rows_values(
+        torch.arange(4096, dtype=torch.int64), 16, 0,
+        reference.INIT_NORMAL, 0.0, 2.0)
+    assert abs(norm.mean().item()) < 0.02
+    assert abs(norm.std().item() - 2.0) < 0.02
+    trunc = reference.init_rows_values(
+        torch.arange(4096, dtype=torch.int64), 16, 0,
+        reference.INIT_TRUNC_NORMAL, 0.0, 1.0)
+    assert trunc.abs().max().item() <= 2.0
+    # deterministic: same ids+seed -> same rows
+    again = reference.init_rows_values(ids, 4, 7, reference.INIT_NORMAL,
+                                       0.0, 1.0)
+    again2 = reference.init_rows_values(ids, 4, 7, reference.INIT_NORMAL,
+                                        0.0, 1.0)
+    assert torch.equal(again, again2)
+
+
+def test_table_with_normal_initializer():
+    table = EmbeddingTable("t", 8, device="cpu", max_rows=100,
+                           initializer=("normal", 0.0, 0.1))
+    rows = table.gather(torch.arange(50, dtype=torch.int64))
+    assert rows.std().item() < 0.2 and rows.std().item() > 0.05
