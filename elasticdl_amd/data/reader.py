@@ -157,6 +157,104 @@ class RecordFileReader(AbstractDataReader):
                 yield f.read(ln)
 
 
+# ----------------------------------------------------------------- recordio
+class RecordIOReader(AbstractDataReader):
+    """Reader for the RecordIO chunk container (data/recordio.py),
+    mirroring the reference's RecordIODataReader
+    (data/reader/recordio_reader.py:27-64): ``data_origin`` is a directory
+    of RecordIO files or a single file; one shard per file named by path
+    with [start, end) record ranges."""
+
+    def __init__(self, data_origin: str, records_per_shard: int = 0):
+        self.data_origin = data_origin
+        self.records_per_shard = records_per_shard
+
+    def _files(self) -> List[str]:
+        if os.path.isdir(self.data_origin):
+            return sorted(
+                os.path.join(self.data_origin, f)
+                for f in os.listdir(self.data_origin)
+                if not f.startswith(".")
+            )
+        return [self.data_origin]
+
+    def create_shards(self) -> List[Tuple[str, int, int]]:
+        from elasticdl_amd.data.recordio import Index
+
+        shards = []
+        for path in self._files():
+            n = Index(path).num_records()
+            step = self.records_per_shard or n
+            shards.extend(
+                (path, lo, min(lo + step, n)) for lo in range(0, n, step)
+            )
+        return shards
+
+    def read_records(self, task: Task) -> Iterator[bytes]:
+        from elasticdl_amd.data.recordio import Scanner
+
+        s = task.shard
+        with Scanner(s.name, s.start, s.end - s.start) as scanner:
+            while True:
+                r = scanner.record()
+                if r is None:
+                    break
+                yield r
+
+
+# -------------------------------------------------------------------- odps
+class ODPSReader(AbstractDataReader):
+    """Offline-testable stub of the reference's ODPS/MaxCompute reader
+    (data/reader/odps_reader.py:51-247). Parses the same
+    ``odps://project/tables/<table>`` origin and env credentials
+    (ODPS_ACCESS_ID/ODPS_ACCESS_KEY/ODPS_ENDPOINT), shards by row ranges,
+    and reads through an injectable client. There is no network in this
+    environment, so the default client refuses with a clear error;
+    production deployments supply one via ``client=`` (any object with
+    ``table_size(project, table) -> int`` and
+    ``read_rows(project, table, start, end) -> iterator``) or use a model
+    zoo custom_data_reader."""
+
+    def __init__(self, data_origin: str, records_per_shard: int = 0,
+                 client=None):
+        if not data_origin.startswith("odps://"):
+            raise ValueError(f"not an ODPS origin: {data_origin!r}")
+        rest = data_origin[len("odps://"):]
+        parts = rest.split("/")
+        if len(parts) < 3 or parts[1] != "tables":
+            raise ValueError(
+                f"expected odps://<project>/tables/<table>, got {data_origin!r}"
+            )
+        self.project, self.table = parts[0], parts[2]
+        self.records_per_shard = records_per_shard
+        self.access_id = os.environ.get("ODPS_ACCESS_ID", "")
+        self.access_key = os.environ.get("ODPS_ACCESS_KEY", "")
+        self.endpoint = os.environ.get("ODPS_ENDPOINT", "")
+        self.client = client
+
+    def _require_client(self):
+        if self.client is None:
+            raise RuntimeError(
+                "ODPS/MaxCompute needs network access, which this "
+                "environment does not have. Inject ODPSReader(client=...) "
+                "or define custom_data_reader() in the model zoo module."
+            )
+        return self.client
+
+    def create_shards(self) -> List[Tuple[str, int, int]]:
+        n = self._require_client().table_size(self.project, self.table)
+        step = self.records_per_shard or n
+        return [
+            (f"{self.project}/{self.table}", lo, min(lo + step, n))
+            for lo in range(0, n, step)
+        ]
+
+    def read_records(self, task: Task) -> Iterator:
+        client = self._require_client()
+        s = task.shard
+        yield from client.read_rows(self.project, self.table, s.start, s.end)
+
+
 # ---------------------------------------------------------------- synthetic
 class SyntheticReader(AbstractDataReader):
     """Deterministic generated records: record i = sample_fn(i)."""
@@ -182,17 +280,30 @@ class SyntheticReader(AbstractDataReader):
 
 
 # ------------------------------------------------------------------ factory
+def _is_recordio(path: str) -> bool:
+    """Sniff the chunk magic (first 4 bytes, little-endian 0x01020304)."""
+    try:
+        with open(path, "rb") as f:
+            return f.read(4) == b"\x04\x03\x02\x01"
+    except OSError:
+        return False
+
+
 def create_data_reader(data_origin: str, records_per_shard: int = 0,
                        **kwargs) -> AbstractDataReader:
     if data_origin.endswith(".csv"):
         return CSVReader(data_origin, records_per_shard, **kwargs)
     if data_origin.endswith((".records", ".edlr")):
         return RecordFileReader(data_origin, records_per_shard)
+    if data_origin.endswith(".recordio"):
+        return RecordIOReader(data_origin, records_per_shard)
     if data_origin.startswith("odps://"):
-        raise NotImplementedError(
-            "ODPS/MaxCompute requires network access; provide a custom "
-            "data reader via the model zoo (custom_data_reader)"
-        )
+        return ODPSReader(data_origin, records_per_shard, **kwargs)
+    if os.path.isdir(data_origin):
+        # directory of RecordIO files (the reference's data_dir layout)
+        return RecordIOReader(data_origin, records_per_shard)
     if os.path.isfile(data_origin):
+        if _is_recordio(data_origin):
+            return RecordIOReader(data_origin, records_per_shard)
         return TextReader(data_origin, records_per_shard, **kwargs)
     raise ValueError(f"cannot infer reader for {data_origin!r}")
