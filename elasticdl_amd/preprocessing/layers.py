@@ -252,6 +252,41 @@ def to_sparse(padded: torch.Tensor, pad_value: int = PAD):
     return torch.sparse_coo_tensor(idx, padded[mask], padded.shape)
 
 
+class ToRagged(nn.Module):
+    """Layer form of ``to_padded`` (reference: to_ragged.py): list-of-
+    lists (or delimiter-joined strings) -> [batch, max_len] padded int64.
+    The padded-with--1 tensor is this framework's ragged representation —
+    every variable-length consumer (EdlEmbedding combiners, ToSparse,
+    ConcatenateWithOffset) reads it."""
+
+    def __init__(self, pad_value: int = PAD, sep: str = ","):
+        super().__init__()
+        self.pad_value = pad_value
+        self.sep = sep
+
+    def forward(self, inputs):
+        if isinstance(inputs, torch.Tensor):
+            return inputs.to(torch.int64)
+        rows = []
+        for row in inputs:
+            if isinstance(row, str):
+                row = [int(v) for v in row.split(self.sep) if v != ""]
+            rows.append(row)
+        return to_padded(rows, self.pad_value)
+
+
+class ToSparse(nn.Module):
+    """Layer form of ``to_sparse`` (reference: to_sparse.py): padded
+    tensor -> torch sparse COO."""
+
+    def __init__(self, pad_value: int = PAD):
+        super().__init__()
+        self.pad_value = pad_value
+
+    def forward(self, inputs: torch.Tensor):
+        return to_sparse(inputs, self.pad_value)
+
+
 def fit_normalizer(values: torch.Tensor) -> "Normalizer":
     """Build a Normalizer from data statistics — the offline-analysis
     stand-in for the reference's analyzer_utils (SQL-computed stats)."""

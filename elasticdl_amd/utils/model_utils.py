@@ -22,6 +22,8 @@ _BUILTIN = {
     "dcn": "elasticdl_amd.models.dcn",
     "cifar10": "elasticdl_amd.models.cifar10",
     "iris": "elasticdl_amd.models.iris",
+    "census_wide_deep": "elasticdl_amd.models.census_wide_deep",
+    "mobilenetv2": "elasticdl_amd.models.mobilenetv2",
 }
 
 
