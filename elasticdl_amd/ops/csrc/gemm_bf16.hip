@@ -159,15 +159,219 @@ __global__ __launch_bounds__(GEMM_THREADS, 2) void gemm_bias_act_kernel(
   }
 }
 
+// ===========================================================================
+// 256x256-tile deep-pipelined kernel (the "phase 2" compute-bound path).
+//
+// The 128x128 2-barrier structure above tops out near ~900 TF on 4096^3+
+// shapes (the guide's measured ceiling for that structure: the vmcnt(0)
+// drain before every __syncthreads stalls the staging queue). This kernel
+// follows the guide's verified 256^2 8-phase template:
+//   - 512 threads = 8 waves (2M x 4N); each wave owns a 128x64 output
+//     sub-tile as 8x4 fragments of v_mfma_f32_16x16x32_bf16;
+//   - LDS 128 KiB: double-buffered K-tiles (A,B each 256x64 bf16 as two
+//     128-row halves), staged with global_load_lds width 16;
+//   - st_16x32 LDS swizzle (physical byte = logical ^ ((logical>>9)&1)<<5
+//     within each half-tile) kills the 16-way ds_read bank conflict of the
+//     linear [128][64] layout (guide: 141x fewer conflicts, +35% at 4k);
+//     applied on the store side by pre-swizzling the per-lane GLOBAL source
+//     (the LDS destination of global_load_lds is wave-uniform+lane*16 and
+//     cannot scatter) and on the read side by XORing the ds_read address;
+//   - phase-interleaved K-loop: 4 phases per K-tile, each = a few
+//     ds_read_b128 fragment loads + raw s_barrier + lgkmcnt(0) +
+//     setprio(1)-wrapped MFMA batch (16) + s_barrier, with the NEXT
+//     K-tile's 8 global_load_lds front-loaded into phase 0 and waited only
+//     ONCE per K-tile (counted issue distance ~= the whole tile's compute,
+//     not the per-tile vmcnt(0)-before-barrier drain of the 128^2 loop);
+//   - prefetch always targets the buffer NOT being read, so correctness
+//     never depends on barrier placement (the template's 3-half-tile-ahead
+//     variant overwrites live slots and is race-prone to reproduce).
+// ===========================================================================
+
+#define BM2 256
+#define BN2 256
+#define BK2 64
+#define GEMM2_THREADS 512
+
+// st_16x32 swizzle of a byte offset within one 16 KiB half-tile
+__device__ inline int swz(int byte) {
+  return byte ^ (((byte >> 9) & 1) << 5);
+}
+
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
+    const __bf16* __restrict__ A,  // [M, K] row-major
+    const __bf16* __restrict__ B,  // [N, K] row-major
+    const float* __restrict__ bias,
+    __bf16* __restrict__ C,  // [M, N] row-major
+    int M, int N, int K) {
+  // [dbuf][operand][half][128*64]
+  __shared__ __bf16 sm[2][2][2][128 * BK2];
+
+  const int ntile_m = (M + BM2 - 1) / BM2;
+  const int ntile_n = (N + BN2 - 1) / BN2;
+  int bid = xcd_swizzle(blockIdx.x, ntile_m * ntile_n);
+  const int bm = bid / ntile_n;
+  const int bn = bid % ntile_n;
+  const int row_a0 = bm * BM2;
+  const int row_b0 = bn * BN2;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;  // 0..7
+  const int wm = wave >> 2;   // 0..1  -> A half = wm
+  const int wn = wave & 3;    // 0..3  -> B half = wn>>1
+
+  // ---- staging: one half-tile = 1024 16B chunks = 2 loads/thread.
+  // dest chunk c -> logical chunk lc (involutive swizzle at 16B grank:
+  // byte=16c, flip of byte-bit5 == flip of chunk-bit1 keyed on chunk-bit5)
+  auto stage_half = [&](int buf, int op, int half, int t_k0) {
+    const __bf16* base = (op == 0 ? A : B);
+    int rows_total = (op == 0 ? M : N);
+    int r0 = (op == 0 ? row_a0 : row_b0) + half * 128;
+#pragma unroll
+    for (int l = 0; l < 2; ++l) {
+      int c = l * 512 + tid;
+      int lc = c ^ (((c >> 5) & 1) << 1);
+      int row = lc >> 3;
+      int k16 = lc & 7;
+      int g = min(r0 + row, rows_total - 1);
+      const __bf16* src = base + (size_t)g * K + t_k0 + k16 * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(
+              &sm[buf][op][half][0] + (l * 512 + wave * 64) * 8),
+          16, 0, 0);
+    }
+  };
+  auto stage_tile = [&](int buf, int t_k0) {
+    stage_half(buf, 0, 0, t_k0);
+    stage_half(buf, 0, 1, t_k0);
+    stage_half(buf, 1, 0, t_k0);
+    stage_half(buf, 1, 1, t_k0);
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int frag_m = lane & 15;
+  const int frag_k = (lane >> 4) * 8;
+
+  // swizzled ds_read of one bf16x8 fragment from a half-tile buffer:
+  // logical byte = row*128 + k*2 (k multiple of 8 -> 16B aligned)
+  auto read_frag = [&](const __bf16* halfbuf, int row, int k) -> bf16x8 {
+    int byte = swz(row * 128 + k * 2);
+    return *reinterpret_cast<const bf16x8*>(
+        reinterpret_cast<const char*>(halfbuf) + byte);
+  };
+
+  const int ktiles = K / BK2;
+  // prologue: tile 0 -> buf 0
+  stage_tile(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __builtin_amdgcn_s_barrier();
+
+  bf16x8 a_frag[8], b_frag[4];
+  for (int t = 0; t < ktiles; ++t) {
+    const int b = t & 1;
+    const __bf16* Ah = &sm[b][0][wm][0];         // this wave's A half
+    const __bf16* Bh0 = &sm[b][1][wn >> 1][0];   // this wave's B half
+    const int arow0 = 0;                          // rows within half
+    const int bcol0 = (wn & 1) * 64;              // cols within half
+
+#pragma unroll
+    for (int phase = 0; phase < 4; ++phase) {
+      const int kk = (phase >> 1) * 32;           // k-chunk of this phase
+      if (phase == 0 && t + 1 < ktiles) {
+        // front-load the next tile's 8 staging ops; waited at loop end
+        stage_tile(b ^ 1, (t + 1) * BK2);
+      }
+      if ((phase & 1) == 0) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          a_frag[i] = read_frag(Ah, arow0 + i * 16 + frag_m, kk + frag_k);
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          b_frag[j] = read_frag(Bh0, bcol0 + j * 16 + frag_m, kk + frag_k);
+      } else {
+#pragma unroll
+        for (int j = 2; j < 4; ++j)
+          b_frag[j] = read_frag(Bh0, bcol0 + j * 16 + frag_m, kk + frag_k);
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)");
+      __builtin_amdgcn_s_setprio(1);
+      const int j0 = (phase & 1) * 2;
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j0 + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j0 + j], acc[i][j0 + j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+    if (t + 1 < ktiles) {
+      asm volatile("s_waitcnt vmcnt(0)");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // ---- epilogue (C/D layout: col = lane&15, row = (lane>>4)*4 + reg)
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row_a0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r;
+        int col = row_b0 + wn * 64 + j * 16 + (lane & 15);
+        if (row < M && col < N) {
+          float v = acc[i][j][r];
+          if (HAS_BIAS) v += bias[col];
+          v = apply_act(v, ACT);
+          C[(size_t)row * N + col] = (__bf16)v;
+        }
+      }
+    }
+  }
+}
+
 extern "C" void edl_gemm_bias_act_bf16(const void* A, const void* B,
                                        const float* bias, void* C, int M,
                                        int N, int K, int act,
                                        hipStream_t stream) {
-  int ntiles = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
-  dim3 grid(ntiles), block(GEMM_THREADS);
   const __bf16* a = reinterpret_cast<const __bf16*>(A);
   const __bf16* b = reinterpret_cast<const __bf16*>(B);
   __bf16* c = reinterpret_cast<__bf16*>(C);
+  // compute-bound shapes take the 256^2 8-phase kernel; tower shapes
+  // (either dim under one 256-tile) keep the 128^2 fused kernel whose
+  // smaller tiles fill the grid and win on memory-bound work
+  const bool big = (M >= 256) && (N >= 256);
+  if (big) {
+    int ntiles = ((M + BM2 - 1) / BM2) * ((N + BN2 - 1) / BN2);
+    dim3 grid(ntiles), block(GEMM2_THREADS);
+#define EDL_GEMM2_CASE(ACTV)                                                  \
+  {                                                                           \
+    if (bias != nullptr)                                                      \
+      gemm_bias_act_256_kernel<ACTV, true>                                    \
+          <<<grid, block, 0, stream>>>(a, b, bias, c, M, N, K);               \
+    else                                                                      \
+      gemm_bias_act_256_kernel<ACTV, false>                                   \
+          <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);            \
+  }
+    switch (act) {
+      case ACT_RELU: EDL_GEMM2_CASE(ACT_RELU); break;
+      case ACT_SIGMOID: EDL_GEMM2_CASE(ACT_SIGMOID); break;
+      default: EDL_GEMM2_CASE(ACT_NONE); break;
+    }
+#undef EDL_GEMM2_CASE
+    return;
+  }
+  int ntiles = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+  dim3 grid(ntiles), block(GEMM_THREADS);
 #define EDL_GEMM_CASE(ACTV)                                                   \
   {                                                                           \
     if (bias != nullptr)                                                      \

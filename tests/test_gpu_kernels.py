@@ -285,7 +285,13 @@ def test_scatter_rows(C):
 
 # ------------------------------- fused GEMM ------------------------------
 @pytest.mark.parametrize(
-    "M,N,K", [(128, 128, 64), (256, 512, 128), (100, 200, 64), (513, 300, 192)]
+    "M,N,K", [
+        (128, 128, 64), (256, 512, 128), (100, 200, 64), (513, 300, 192),
+        # 256^2 8-phase path (M>=256 and N>=256), incl. masked edges and
+        # multi-K-tile pipelining
+        (256, 256, 64), (256, 256, 128), (512, 768, 256), (300, 257, 192),
+        (1024, 512, 1024),
+    ]
 )
 def test_gemm_matches_matmul(C, M, N, K):
     torch.manual_seed(0)
@@ -309,6 +315,27 @@ def test_gemm_bias_relu(C):
     torch.cuda.synchronize()
     ref = torch.relu(a.float() @ b.float().t() + bias)
     assert torch.allclose(out.float(), ref, atol=2e-1, rtol=2e-2)
+
+
+def test_gemm_256_path_bias_relu_race_screen(C):
+    """256^2 8-phase kernel: fused epilogue + repeatability (multi-run
+    race screen per the guide's discipline for new sync structures)."""
+    torch.manual_seed(2)
+    a = torch.randn(512, 256, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(384, 256, dtype=torch.bfloat16, device="cuda")
+    bias = torch.randn(384, dtype=torch.float32, device="cuda")
+    ref = torch.relu(a.float() @ b.float().t() + bias)
+    first = None
+    for _ in range(5):
+        out = C.gemm_bias_act(a, b, bias, 1)
+        torch.cuda.synchronize()
+        assert torch.allclose(out.float(), ref, atol=2e-1, rtol=2e-2), (
+            (out.float() - ref).abs().max()
+        )
+        if first is None:
+            first = out.clone()
+        else:
+            assert torch.equal(out, first)  # bitwise repeatable
 
 
 def test_engine_gpu_matches_cpu_end_to_end():
