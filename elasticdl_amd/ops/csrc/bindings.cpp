@@ -73,6 +73,8 @@ void edl_scatter_rows(float*, const int32_t*, const float*, int64_t, int64_t,
                       hipStream_t);
 void edl_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
                             int, int, int, int, hipStream_t);
+void edl_gemm256_variant_bf16(const void*, const void*, void*, int, int,
+                              int, int, hipStream_t);
 void edl_fused_sgd_bf16(void*, float*, float*, const void*, int64_t, float,
                         float, bool, float, float, hipStream_t);
 void edl_fused_adamw_bf16(void*, float*, float*, float*, const void*, int64_t,
@@ -401,6 +403,17 @@ torch::Tensor gemm_bias_act(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+// benchmark-only: 256^2 kernel with explicit barrier-variant selection
+torch::Tensor gemm256_bench(torch::Tensor a, torch::Tensor b, int64_t bars) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.size(1) == b.size(1) && a.size(1) % 64 == 0);
+  auto c = torch::empty({a.size(0), b.size(0)}, a.options());
+  edl_gemm256_variant_bf16(a.data_ptr(), b.data_ptr(), c.data_ptr(),
+                           a.size(0), b.size(0), a.size(1),
+                           static_cast<int>(bars), cur_stream());
+  return c;
+}
+
 // ---------------------- worker-side fused optimizers --------------------
 void fused_sgd_bf16(torch::Tensor p, torch::Tensor master, torch::Tensor vel,
                     torch::Tensor g, double lr, double mu, bool nesterov,
@@ -458,6 +471,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_rows", &gather_rows);
   m.def("scatter_rows", &scatter_rows);
   m.def("gemm_bias_act", &gemm_bias_act);
+  m.def("gemm256_bench", &gemm256_bench);
   m.def("fused_sgd_bf16", &fused_sgd_bf16);
   m.def("fused_adamw_bf16", &fused_adamw_bf16);
 }
