@@ -197,11 +197,15 @@ __device__ inline int swz(int byte) {
   return byte ^ (((byte >> 9) & 1) << 5);
 }
 
-// BARS: phase-synchronization structure (A/B-tested on hardware):
-//   0 = free-running waves (only the buffer-swap barrier),
-//   1 = one barrier before each MFMA batch,
-//   2 = the template's two-barrier lockstep.
-template <int ACT, bool HAS_BIAS, int BARS = 2>
+// BARS: phase-synchronization structure (A/B-tested on hardware; measured
+// 4096^3/8192^3 bf16, ab_gemm1/2):
+//   0 = pure free-run: no phase barriers, no explicit waits, compiler
+//       schedules counted lgkmcnt itself
+//   1 = free-run + per-phase lgkmcnt(0) + setprio around MFMA  [1041/1125 TF]
+//   2 = one barrier before each MFMA batch                      [987/1039]
+//   3 = the template's two-barrier lockstep                     [859/941]
+//   4 = whole-tile fragment reads then whole-tile MFMA (no phases)
+template <int ACT, bool HAS_BIAS, int BARS = 1>
 __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
     const __bf16* __restrict__ A,  // [M, K] row-major
     const __bf16* __restrict__ B,  // [N, K] row-major
@@ -285,37 +289,65 @@ __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
     const int arow0 = 0;                          // rows within half
     const int bcol0 = (wn & 1) * 64;              // cols within half
 
+    if (BARS == 4) {
+      // whole-tile batch: all fragment reads, then all MFMAs; the
+      // compiler interleaves with its own counted lgkmcnt waits
+      if (t + 1 < ktiles) stage_tile(b ^ 1, (t + 1) * BK2);
+      bf16x8 a2[2][8], b2[2][4];
 #pragma unroll
-    for (int phase = 0; phase < 4; ++phase) {
-      const int kk = (phase >> 1) * 32;           // k-chunk of this phase
-      if (phase == 0 && t + 1 < ktiles) {
-        // front-load the next tile's 8 staging ops; waited at loop end
-        stage_tile(b ^ 1, (t + 1) * BK2);
-      }
-      if ((phase & 1) == 0) {
+      for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
         for (int i = 0; i < 8; ++i)
-          a_frag[i] = read_frag(Ah, arow0 + i * 16 + frag_m, kk + frag_k);
+          a2[kk][i] =
+              read_frag(Ah, i * 16 + frag_m, kk * 32 + frag_k);
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
-          b_frag[j] = read_frag(Bh0, bcol0 + j * 16 + frag_m, kk + frag_k);
-      } else {
-#pragma unroll
-        for (int j = 2; j < 4; ++j)
-          b_frag[j] = read_frag(Bh0, bcol0 + j * 16 + frag_m, kk + frag_k);
+        for (int j = 0; j < 4; ++j)
+          b2[kk][j] =
+              read_frag(Bh0, bcol0 + j * 16 + frag_m, kk * 32 + frag_k);
       }
-      if (BARS >= 1) __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)");
-      __builtin_amdgcn_s_setprio(1);
-      const int j0 = (phase & 1) * 2;
 #pragma unroll
-      for (int i = 0; i < 8; ++i)
+      for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
-          acc[i][j0 + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[i], b_frag[j0 + j], acc[i][j0 + j], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-      if (BARS >= 2) __builtin_amdgcn_s_barrier();
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a2[kk][i], b2[kk][j], acc[i][j], 0, 0, 0);
+    } else {
+#pragma unroll
+      for (int phase = 0; phase < 4; ++phase) {
+        const int kk = (phase >> 1) * 32;  // k-chunk of this phase
+        if (phase == 0 && t + 1 < ktiles) {
+          // front-load the next tile's 8 staging ops; waited at loop end
+          stage_tile(b ^ 1, (t + 1) * BK2);
+        }
+        if ((phase & 1) == 0) {
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            a_frag[i] = read_frag(Ah, arow0 + i * 16 + frag_m, kk + frag_k);
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            b_frag[j] = read_frag(Bh0, bcol0 + j * 16 + frag_m, kk + frag_k);
+        } else {
+#pragma unroll
+          for (int j = 2; j < 4; ++j)
+            b_frag[j] = read_frag(Bh0, bcol0 + j * 16 + frag_m, kk + frag_k);
+        }
+        if (BARS >= 2) __builtin_amdgcn_s_barrier();
+        if (BARS >= 1) {
+          asm volatile("s_waitcnt lgkmcnt(0)");
+          __builtin_amdgcn_s_setprio(1);
+        }
+        const int j0 = (phase & 1) * 2;
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j0 + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[i], b_frag[j0 + j], acc[i][j0 + j], 0, 0, 0);
+        if (BARS >= 1) __builtin_amdgcn_s_setprio(0);
+        if (BARS >= 3) __builtin_amdgcn_s_barrier();
+      }
     }
     if (t + 1 < ktiles) {
       asm volatile("s_waitcnt vmcnt(0)");
@@ -361,8 +393,16 @@ extern "C" void edl_gemm256_variant_bf16(const void* A, const void* B,
       gemm_bias_act_256_kernel<ACT_NONE, false, 1>
           <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
       break;
-    default:
+    case 2:
       gemm_bias_act_256_kernel<ACT_NONE, false, 2>
+          <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
+      break;
+    case 4:
+      gemm_bias_act_256_kernel<ACT_NONE, false, 4>
+          <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
+      break;
+    default:
+      gemm_bias_act_256_kernel<ACT_NONE, false, 3>
           <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
   }
 }
