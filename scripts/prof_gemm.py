@@ -22,17 +22,26 @@ def main():
     k = int(sys.argv[3]) if len(sys.argv) > 3 else n
     iters = int(os.environ.get("ITERS", "30"))
     lib = os.environ.get("LIB") == "1"
+    bars = os.environ.get("BARS")
     C = require_native()
     a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
     b = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+
+    def run():
+        if lib:
+            return a @ b.t()
+        if bars is not None:
+            return C.gemm256_bench(a, b, int(bars))
+        return C.gemm_bias_act(a, b, None, 0)
+
     for _ in range(5):
-        out = (a @ b.t()) if lib else C.gemm_bias_act(a, b, None, 0)
+        out = run()
     torch.cuda.synchronize()
     import time
 
     t0 = time.perf_counter()
     for _ in range(iters):
-        out = (a @ b.t()) if lib else C.gemm_bias_act(a, b, None, 0)
+        out = run()
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
     tf = 2.0 * m * n * k / dt / 1e12
