@@ -192,9 +192,28 @@ __global__ __launch_bounds__(GEMM_THREADS, 2) void gemm_bias_act_kernel(
 #define BK2 64
 #define GEMM2_THREADS 512
 
-// st_16x32 swizzle of a byte offset within one 16 KiB half-tile
-__device__ inline int swz(int byte) {
-  return byte ^ (((byte >> 9) & 1) << 5);
+// LDS half-tile layout: contiguous 16x32 subtiles (1024 B each), i.e.
+// logical byte of element (row, k) in a [128][64] bf16 half-tile is
+//   ((row>>4)*2 + (k>>5))*1024 + (row&15)*64 + (k&31)*2
+// so a fragment-read row stride is 64 B (16 banks) instead of the
+// row-major 128 B (32 banks, which aliases 16 lanes onto 2 bank groups
+// and defeats the swizzle). The st_16x32 swizzle then XORs byte-bit5
+// with byte-bit9 *within* each subtile (rows 8-15 shift 32 B), spreading
+// a 16-lane fragment column read across 8 bank groups (~2-way).
+__device__ inline int lds_byte(int row, int k) {
+  int lb = (((row >> 4) << 1) + (k >> 5)) * 1024 + (row & 15) * 64 +
+           (k & 31) * 2;
+  return lb ^ (((lb >> 9) & 1) << 5);
+}
+
+// inverse for the store side: which logical (row, k) lives at physical
+// 16 B chunk c (the global_load_lds destination is linear in lane)
+__device__ inline void lds_chunk_src(int c, int& row, int& k) {
+  int lb = (c * 16) ^ ((((c * 16) >> 9) & 1) << 5);
+  int subtile = lb >> 10;
+  int within = lb & 1023;
+  row = ((subtile >> 1) << 4) + ((within >> 6) & 15);
+  k = ((subtile & 1) << 5) + ((within & 63) >> 1);
 }
 
 // BARS: phase-synchronization structure (A/B-tested on hardware; measured
@@ -229,9 +248,10 @@ __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
   const int wm = wave >> 2;   // 0..1  -> A half = wm
   const int wn = wave & 3;    // 0..3  -> B half = wn>>1
 
-  // ---- staging: one half-tile = 1024 16B chunks = 2 loads/thread.
-  // dest chunk c -> logical chunk lc (involutive swizzle at 16B grank:
-  // byte=16c, flip of byte-bit5 == flip of chunk-bit1 keyed on chunk-bit5)
+  // ---- staging: one half-tile = 1024 16B chunks = 2 loads/thread. The
+  // LDS destination is linear (wave-uniform base + lane*16); the swizzled
+  // subtile layout is realized by pre-swizzling the per-lane GLOBAL
+  // source: chunk c fetches the (row, k) that lds_byte maps to 16c.
   auto stage_half = [&](int buf, int op, int half, int t_k0) {
     const __bf16* base = (op == 0 ? A : B);
     int rows_total = (op == 0 ? M : N);
@@ -239,11 +259,10 @@ __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
 #pragma unroll
     for (int l = 0; l < 2; ++l) {
       int c = l * 512 + tid;
-      int lc = c ^ (((c >> 5) & 1) << 1);
-      int row = lc >> 3;
-      int k16 = lc & 7;
+      int row, k;
+      lds_chunk_src(c, row, k);
       int g = min(r0 + row, rows_total - 1);
-      const __bf16* src = base + (size_t)g * K + t_k0 + k16 * 8;
+      const __bf16* src = base + (size_t)g * K + t_k0 + k;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(
@@ -267,12 +286,11 @@ __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
   const int frag_m = lane & 15;
   const int frag_k = (lane >> 4) * 8;
 
-  // swizzled ds_read of one bf16x8 fragment from a half-tile buffer:
-  // logical byte = row*128 + k*2 (k multiple of 8 -> 16B aligned)
+  // swizzled ds_read of one bf16x8 fragment (k multiple of 8, within one
+  // 32-col subtile -> 16B aligned physical address)
   auto read_frag = [&](const __bf16* halfbuf, int row, int k) -> bf16x8 {
-    int byte = swz(row * 128 + k * 2);
     return *reinterpret_cast<const bf16x8*>(
-        reinterpret_cast<const char*>(halfbuf) + byte);
+        reinterpret_cast<const char*>(halfbuf) + lds_byte(row, k));
   };
 
   const int ktiles = K / BK2;
