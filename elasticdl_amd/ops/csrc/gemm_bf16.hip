@@ -299,13 +299,57 @@ __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
   asm volatile("s_waitcnt vmcnt(0)");
   __builtin_amdgcn_s_barrier();
 
+  const int bcol0_w = (wn & 1) * 64;  // cols within this wave's B half
+
+  if (BARS == 5) {
+    // software-pipelined at k-chunk granularity: two fragment register
+    // sets; the ds_reads for the NEXT 32-deep k-chunk (or the next
+    // K-tile's first chunk, after the staging wait) are issued before
+    // the current chunk's 32-MFMA batch, so LDS latency and the
+    // buffer-swap wait hide under MFMA issue.
+    bf16x8 aA[8], bA[4], aB[8], bB[4];
+    auto read_set = [&](bf16x8* af, bf16x8* bf, int buf, int kk) {
+      const __bf16* Ah = &sm[buf][0][wm][0];
+      const __bf16* Bh = &sm[buf][1][wn >> 1][0];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        af[i] = read_frag(Ah, i * 16 + frag_m, kk + frag_k);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = read_frag(Bh, bcol0_w + j * 16 + frag_m, kk + frag_k);
+    };
+    auto mfma_set = [&](const bf16x8* af, const bf16x8* bf) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    };
+    read_set(aA, bA, 0, 0);
+    for (int t = 0; t < ktiles; ++t) {
+      const int b = t & 1;
+      if (t + 1 < ktiles) stage_tile(b ^ 1, (t + 1) * BK2);
+      read_set(aB, bB, b, 32);
+      mfma_set(aA, bA);
+      if (t + 1 < ktiles) {
+        asm volatile("s_waitcnt vmcnt(0)");
+        __builtin_amdgcn_s_barrier();
+        read_set(aA, bA, b ^ 1, 0);
+      }
+      mfma_set(aB, bB);
+    }
+    goto epilogue;
+  }
+
+  {
   bf16x8 a_frag[8], b_frag[4];
   for (int t = 0; t < ktiles; ++t) {
     const int b = t & 1;
     const __bf16* Ah = &sm[b][0][wm][0];         // this wave's A half
     const __bf16* Bh0 = &sm[b][1][wn >> 1][0];   // this wave's B half
     const int arow0 = 0;                          // rows within half
-    const int bcol0 = (wn & 1) * 64;              // cols within half
+    const int bcol0 = bcol0_w;                    // cols within half
 
     if (BARS == 4) {
       // whole-tile batch: all fragment reads, then all MFMAs; the
@@ -373,6 +417,9 @@ __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
     }
   }
 
+  }  // non-pipelined variants
+
+epilogue:
   // ---- epilogue (C/D layout: col = lane&15, row = (lane>>4)*4 + reg)
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
@@ -417,6 +464,10 @@ extern "C" void edl_gemm256_variant_bf16(const void* A, const void* B,
       break;
     case 4:
       gemm_bias_act_256_kernel<ACT_NONE, false, 4>
+          <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
+      break;
+    case 5:
+      gemm_bias_act_256_kernel<ACT_NONE, false, 5>
           <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
       break;
     default:
