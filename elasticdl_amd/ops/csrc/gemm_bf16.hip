@@ -440,6 +440,138 @@ epilogue:
   }
 }
 
+// ===========================================================================
+// BK=32 4-slot ring variant: the never-drain vmcnt pipeline. Four 32 KiB
+// K-tiles live in LDS; every iteration waits s_waitcnt vmcnt(8) (its own
+// tile's 4 staging loads done, the next TWO tiles' 8 loads still in
+// flight) — the staging queue is never drained, which is the structural
+// fix the guide identifies for the ~900 TF 2-barrier ceiling (AITER's
+// interleaved MFMA<->buffer_load with vmcnt(2/5/12), never 0).
+// Safety: the slot overwritten by tile t+3's staging is tile t-1's,
+// whose ds_reads completed before every wave crossed this iteration's
+// barrier (reads are consumed by MFMAs that precede the barrier in
+// program order). Requires K >= 4*32 (dispatcher guards).
+// ===========================================================================
+
+__device__ inline int lds_byte32(int row, int k) {
+  // [128][32] half-tile as contiguous 16x32 subtiles + st_16x32 swizzle
+  int lb = ((row >> 4) << 10) + (row & 15) * 64 + k * 2;
+  return lb ^ (((lb >> 9) & 1) << 5);
+}
+
+template <int ACT, bool HAS_BIAS>
+__global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_bk32_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    const float* __restrict__ bias, __bf16* __restrict__ C, int M, int N,
+    int K) {
+  __shared__ __bf16 sm[4][2][2][128 * 32];  // [slot][op][half][8 KiB]
+
+  const int ntile_m = (M + BM2 - 1) / BM2;
+  const int ntile_n = (N + BN2 - 1) / BN2;
+  int bid = xcd_swizzle(blockIdx.x, ntile_m * ntile_n);
+  const int bm = bid / ntile_n;
+  const int bn = bid % ntile_n;
+  const int row_a0 = bm * BM2;
+  const int row_b0 = bn * BN2;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  const int bcol0 = (wn & 1) * 64;
+
+  // one half-tile (128x32 = 512 chunks) = 1 load/thread
+  auto stage_half = [&](int slot, int op, int half, int t_k0) {
+    const __bf16* base = (op == 0 ? A : B);
+    int rows_total = (op == 0 ? M : N);
+    int r0 = (op == 0 ? row_a0 : row_b0) + half * 128;
+    int c = tid;
+    int lb = (c * 16) ^ ((((c * 16) >> 9) & 1) << 5);
+    int row = ((lb >> 10) << 4) + ((lb >> 6) & 15);
+    int k = (lb & 63) >> 1;
+    int g = min(r0 + row, rows_total - 1);
+    const __bf16* src = base + (size_t)g * K + t_k0 + k;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(
+            &sm[slot][op][half][0] + wave * 64 * 8),
+        16, 0, 0);
+  };
+  auto stage_tile = [&](int slot, int t_k0) {
+    stage_half(slot, 0, 0, t_k0);
+    stage_half(slot, 0, 1, t_k0);
+    stage_half(slot, 1, 0, t_k0);
+    stage_half(slot, 1, 1, t_k0);
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int frag_m = lane & 15;
+  const int frag_k = (lane >> 4) * 8;
+  auto read_frag32 = [&](const __bf16* halfbuf, int row, int k) -> bf16x8 {
+    return *reinterpret_cast<const bf16x8*>(
+        reinterpret_cast<const char*>(halfbuf) + lds_byte32(row, k));
+  };
+
+  const int ktiles = K / 32;  // caller guarantees >= 4
+  stage_tile(0, 0);
+  stage_tile(1, 32);
+  stage_tile(2, 64);
+
+  bf16x8 a_frag[8], b_frag[4];
+  for (int t = 0; t < ktiles; ++t) {
+    const int slot = t & 3;
+    // own tile complete; the next two tiles' 8 loads stay in flight
+    if (t + 2 < ktiles) {
+      asm volatile("s_waitcnt vmcnt(8)");
+    } else if (t + 1 < ktiles) {
+      asm volatile("s_waitcnt vmcnt(4)");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)");
+    }
+    __builtin_amdgcn_s_barrier();
+    if (t + 3 < ktiles) stage_tile((t + 3) & 3, (t + 3) * 32);
+
+    const __bf16* Ah = &sm[slot][0][wm][0];
+    const __bf16* Bh = &sm[slot][1][wn >> 1][0];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      a_frag[i] = read_frag32(Ah, i * 16 + frag_m, frag_k);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      b_frag[j] = read_frag32(Bh, bcol0 + j * 16 + frag_m, frag_k);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row_a0 + wm * 128 + i * 16 + (lane >> 4) * 4 + r;
+        int col = row_b0 + wn * 64 + j * 16 + (lane & 15);
+        if (row < M && col < N) {
+          float v = acc[i][j][r];
+          if (HAS_BIAS) v += bias[col];
+          v = apply_act(v, ACT);
+          C[(size_t)row * N + col] = (__bf16)v;
+        }
+      }
+    }
+  }
+}
+
 // benchmark-only entry: pick the barrier variant explicitly
 extern "C" void edl_gemm256_variant_bf16(const void* A, const void* B,
                                          void* C, int M, int N, int K,
@@ -468,6 +600,10 @@ extern "C" void edl_gemm256_variant_bf16(const void* A, const void* B,
       break;
     case 5:
       gemm_bias_act_256_kernel<ACT_NONE, false, 5>
+          <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
+      break;
+    case 6:
+      gemm_bias_act_bk32_kernel<ACT_NONE, false>
           <<<grid, block, 0, stream>>>(a, b, nullptr, c, M, N, K);
       break;
     default:

@@ -37,7 +37,7 @@ def main():
         ref = None
         t_lib = timeit(lambda: a @ b.t())
         print(f"N={n} hipBLASLt: {flops/t_lib/1e12:.0f} TF")
-        for bars in (5, 4, 1, 0):
+        for bars in (6, 1, 0):
             out = C.gemm256_bench(a, b, bars)
             torch.cuda.synchronize()
             if ref is None:
