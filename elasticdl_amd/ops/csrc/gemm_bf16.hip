@@ -224,7 +224,11 @@ __device__ inline void lds_chunk_src(int c, int& row, int& k) {
 //   2 = one barrier before each MFMA batch                      [987/1039]
 //   3 = the template's two-barrier lockstep                     [859/941]
 //   4 = whole-tile fragment reads then whole-tile MFMA (no phases)
-template <int ACT, bool HAS_BIAS, int BARS = 1>
+//   5 = k-chunk software pipelining with two register sets (spills; slow)
+// Production default is 0 — the full hardware A/B ladder is recorded in
+// profiles/kernels_r02.md (free-run > +asm > barrier variants > BK=32
+// vmcnt ring, on this structure).
+template <int ACT, bool HAS_BIAS, int BARS = 0>
 __global__ __launch_bounds__(GEMM2_THREADS, 1) void gemm_bias_act_256_kernel(
     const __bf16* __restrict__ A,  // [M, K] row-major
     const __bf16* __restrict__ B,  // [N, K] row-major
