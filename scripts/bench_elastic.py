@@ -35,7 +35,10 @@ def main():
     ap.add_argument("--rejoin", action="store_true", default=True)
     args = ap.parse_args()
 
-    tmp = tempfile.mkdtemp(prefix="edl-elastic-")
+    tmp = os.environ.get("EDL_ELASTIC_TMP") or tempfile.mkdtemp(
+        prefix="edl-elastic-"
+    )
+    os.makedirs(tmp, exist_ok=True)
     cmd = [
         sys.executable, "-m", "elasticdl_amd.master.main",
         "--model_def", args.model,
