@@ -47,4 +47,7 @@ def collate_fn(records):
 
 def feed(batch, device, dtype=None):
     x, y = batch
-    return x.to(device), y.to(device)
+    x = x.to(device)
+    if dtype is not None:
+        x = x.to(dtype)
+    return x, y.to(device)

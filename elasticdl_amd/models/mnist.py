@@ -49,7 +49,10 @@ def eval_metrics_fn():
 
 def feed(batch, device, dtype=None):
     images, labels = batch
-    return images.to(device), labels.to(device)
+    images = images.to(device)
+    if dtype is not None:
+        images = images.to(dtype)
+    return images, labels.to(device)
 
 
 def custom_data_reader(data_origin: str = ""):

@@ -167,9 +167,19 @@ class AllReduceTrainer(Trainer):
     def _feed(self, batch):
         if self.spec.feed_fn is not None:
             try:
-                return self.spec.feed_fn(batch, self.device, self.dtype)
+                x, y = self.spec.feed_fn(batch, self.device, self.dtype)
             except TypeError:
-                return self.spec.feed_fn(batch, self.device)
+                x, y = self.spec.feed_fn(batch, self.device)
+            # a zoo feed() that ignores dtype would make every GPU
+            # minibatch fail against the bf16 model (and the retry loop
+            # turns that into a silent 64x slowdown) — cast defensively
+            if (
+                isinstance(x, torch.Tensor)
+                and x.is_floating_point()
+                and x.dtype != self.dtype
+            ):
+                x = x.to(self.dtype)
+            return x, y
         x, y = batch
         return x.to(self.device, self.dtype), y.to(self.device)
 

@@ -127,3 +127,43 @@ def test_sharded_engine_world1_gpu():
     rows = eng.pull_embedding_vectors("e", torch.arange(100))
     torch.cuda.synchronize()
     assert rows.shape == (100, 16)
+
+
+def test_mnist_allreduce_trainer_gpu_minibatch():
+    """Regression: the elastic bench's mnist path on GPU (bf16 model)
+    must train a minibatch on the FIRST try — a zoo feed() that drops the
+    dtype made every minibatch fail and burn the 64-retry loop."""
+    from elasticdl_amd.master.rendezvous import ElasticRendezvousServer
+    from elasticdl_amd.utils.model_utils import get_model_spec
+    from elasticdl_amd.worker.allreduce_trainer import AllReduceTrainer
+
+    rdzv = ElasticRendezvousServer("127.0.0.1")
+    rdzv.start()
+    rdzv._flip_delay_sec = 0.0
+    rdzv.add_worker("w0")
+
+    class MC:
+        worker_host = "w0"
+
+        def get_comm_rank(self, host):
+            return rdzv.get_comm_rank(host)
+
+        def rendezvous_addr(self, info):
+            return "127.0.0.1", info["rendezvous_port"]
+
+        def report_training_loop_status(self, s):
+            pass
+
+    import os
+
+    os.environ["EDL_BACKEND"] = "gloo"
+    try:
+        tr = AllReduceTrainer(get_model_spec("mnist"), MC(), device="cuda")
+        assert tr.dtype == torch.bfloat16
+        x = torch.rand(16, 1, 28, 28)
+        y = torch.randint(0, 10, (16,))
+        loss, version = tr.train_minibatch((x, y))
+        assert torch.isfinite(loss) and version == 1
+        tr.comm.teardown()
+    finally:
+        os.environ.pop("EDL_BACKEND", None)
