@@ -72,10 +72,21 @@ class AllReduceTrainer(Trainer):
                 and self.comm.rendezvous_id >= 0 and self.comm.world_size > 0:
             return
         self._last_world_check = now
-        reformed = self.comm.ensure_communicator()
-        if reformed or self.comm.need_broadcast:
-            self._broadcast_state()
-            self.comm.need_broadcast = False
+        try:
+            reformed = self.comm.ensure_communicator()
+            if reformed or self.comm.need_broadcast:
+                self._broadcast_state()
+                self.comm.need_broadcast = False
+        except RuntimeError as e:
+            # joining a world that dissolved while we bootstrapped (a peer
+            # exited between staging and init_process_group) — poison the
+            # generation and re-raise as a collective error so the
+            # minibatch retry loop re-rendezvouses instead of crashing
+            if not is_collective_error(e):
+                raise
+            logger.warning("communicator init failed (%s); will retry", e)
+            self.comm.handle_collective_failure()
+            raise
         self._adjust_accumulation()
 
     def _broadcast_state(self) -> None:
@@ -219,9 +230,15 @@ class AllReduceTrainer(Trainer):
         # present) is unchanged when unset.
         min_world = int(os.environ.get("EDL_MIN_WORLD", "0"))
         if min_world > 1:
-            deadline = time.monotonic() + 120
+            deadline = time.monotonic() + float(
+                os.environ.get("EDL_MIN_WORLD_TIMEOUT_SEC", "60")
+            )
             while time.monotonic() < deadline:
-                self.init_communicator_if_needed(force=True)
+                try:
+                    self.init_communicator_if_needed(force=True)
+                except RuntimeError as e:  # dissolved world: retry below
+                    if not is_collective_error(e):
+                        raise
                 if self.comm.world_size >= min_world:
                     break
                 time.sleep(1)
