@@ -126,7 +126,7 @@ def gpu_pass():
     for it in range(50):
         # duplicate-heavy id stream incl. boundary ids
         ids = torch.cat([
-            torch.randint(0, 1 << 20, (2000,), generator=g),
+            torch.randint(0, 40000, (2000,), generator=g),
             torch.tensor([0, 0, 0, (1 << 62), (1 << 62) + 1]),
         ])
         rows = eng.pull_embedding_vectors("emb", ids)
@@ -141,6 +141,17 @@ def gpu_pass():
         )
     for t in eng.tables.values():
         t.check_health()
+    guards.append(_canary())
+
+    # overflow must be DETECTED (error flag), never silently corrupt
+    tiny = PSEngine(device="cuda", embedding_max_rows=64)
+    tiny.push_model({}, [{"name": "t", "dim": 8}])
+    tiny.pull_embedding_vectors("t", torch.arange(4096, dtype=torch.int64))
+    try:
+        tiny.tables["t"].check_health()
+        raise AssertionError("overflow not detected")
+    except RuntimeError as e:
+        assert "overflow" in str(e)
     guards.append(_canary())
 
     # GEMM kernels at masked edge shapes (row/col tails probe OOB writes)
