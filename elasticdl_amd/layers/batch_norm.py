@@ -1,0 +1,122 @@
+"""Fused channels_last BatchNorm for MI355X.
+
+torch's native channels_last batch-norm kernels are the dominant cost of
+the ResNet50 training step on MI355X (63% of steady-state GPU time,
+profiles/resnet_r02.md) while BN is purely memory-bound. This module is a
+drop-in ``nn.BatchNorm2d`` subclass that routes the bf16 + channels_last
++ CUDA training path through the hand-written NHWC kernels
+(ops/csrc/bn_kernels.hip): register-accumulated per-channel statistics,
+single-pass vec8 normalize, two-kernel backward. Every other
+configuration (CPU, fp32, eval without the extension, odd channel
+counts) falls back to the stock implementation, so state_dict layout,
+running-stats semantics and numerics contracts are unchanged.
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+
+def _flat_nhwc(x: torch.Tensor) -> torch.Tensor:
+    """channels_last [N,C,H,W] -> [N*H*W, C] view (no copy)."""
+    n, c, h, w = x.shape
+    return x.permute(0, 2, 3, 1).reshape(n * h * w, c)
+
+
+def _supported(x: torch.Tensor) -> bool:
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4):
+        return False
+    if not x.is_contiguous(memory_format=torch.channels_last):
+        return False
+    c = x.shape[1]
+    return c >= 8 and c % 8 == 0 and c <= 2048 and 256 % (c // 8) == 0
+
+
+class _FusedBNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, mean, rstd):
+        from elasticdl_amd.ops import require_native
+
+        C = require_native()
+        xf = _flat_nhwc(x)
+        y = C.bn_apply(xf, mean, rstd, weight.float(), bias.float(), False)
+        ctx.save_for_backward(xf, mean, rstd, weight)
+        ctx.shape = x.shape
+        return y.reshape(x.shape[0], x.shape[2], x.shape[3],
+                         x.shape[1]).permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        from elasticdl_amd.ops import require_native
+
+        C = require_native()
+        xf, mean, rstd, weight = ctx.saved_tensors
+        dyf = _flat_nhwc(
+            dy.contiguous(memory_format=torch.channels_last)
+        ).contiguous()
+        s1, s2 = C.bn_bwd_reduce(xf, dyf, None, mean, rstd)
+        r = xf.shape[0]
+        g = weight.float()
+        # dx = a*dy - b*x + c with per-channel coefficients:
+        #   a = g*rstd ; b = g*rstd^2*s2/R ; c = g*rstd*(mean*rstd*s2 - s1)/R
+        a = g * rstd
+        b = a * rstd * s2 / r
+        c = a * (mean * rstd * s2 - s1) / r
+        dxf = C.bn_bwd_apply(xf, dyf, None, a, b, c)
+        n, ch, h, w = ctx.shape
+        dx = dxf.reshape(n, h, w, ch).permute(0, 3, 1, 2)
+        dweight = s2.to(weight.dtype)
+        dbias = s1.to(weight.dtype)
+        return dx, dweight, dbias, None, None
+
+
+class FusedBatchNorm2d(nn.BatchNorm2d):
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from elasticdl_amd.ops import use_native
+
+        if not (self.training and self.affine and _supported(x)
+                and use_native(x.device)):
+            return super().forward(x)
+
+        from elasticdl_amd.ops import require_native
+
+        C = require_native()
+        xf = _flat_nhwc(x)
+        r = xf.shape[0]
+        sum_, sumsq = C.bn_stats(xf)
+        mean = sum_ / r
+        var = (sumsq / r - mean * mean).clamp_(min=0)
+        rstd = (var + self.eps).rsqrt()
+
+        if self.track_running_stats and self.running_mean is not None:
+            with torch.no_grad():
+                m = self.momentum if self.momentum is not None else 0.1
+                unbias = var * (r / max(r - 1, 1))
+                self.running_mean.mul_(1 - m).add_(mean, alpha=m)
+                self.running_var.mul_(1 - m).add_(unbias, alpha=m)
+                if self.num_batches_tracked is not None:
+                    self.num_batches_tracked += 1
+
+        return _FusedBNFn.apply(x, self.weight, self.bias, mean, rstd)
+
+
+def convert_to_fused_bn(module: nn.Module) -> nn.Module:
+    """Swap every nn.BatchNorm2d in a model for FusedBatchNorm2d
+    (parameters/buffers carried over; state_dict-compatible)."""
+    for name, child in module.named_children():
+        if type(child) is nn.BatchNorm2d:
+            fused = FusedBatchNorm2d(
+                child.num_features, eps=child.eps, momentum=child.momentum,
+                affine=child.affine,
+                track_running_stats=child.track_running_stats,
+            )
+            fused.load_state_dict(child.state_dict())
+            fused = fused.to(
+                device=child.weight.device if child.affine else "cpu",
+                dtype=child.weight.dtype if child.affine else None,
+            )
+            setattr(module, name, fused)
+        else:
+            convert_to_fused_bn(child)
+    return module

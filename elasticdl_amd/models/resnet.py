@@ -82,8 +82,16 @@ class ResNet(nn.Module):
         return self.fc(x)
 
 
-def resnet50(num_classes: int = 1000) -> ResNet:
-    return ResNet([3, 4, 6, 3], num_classes)
+def resnet50(num_classes: int = 1000, fused_bn: bool = True) -> ResNet:
+    model = ResNet([3, 4, 6, 3], num_classes)
+    if fused_bn:
+        # MI355X-native channels_last BN kernels (torch's are the
+        # dominant step cost — profiles/resnet_r02.md); a transparent
+        # fallback keeps CPU/fp32 paths on stock BN
+        from elasticdl_amd.layers.batch_norm import convert_to_fused_bn
+
+        convert_to_fused_bn(model)
+    return model
 
 
 def resnet18_cifar(num_classes: int = 10) -> ResNet:

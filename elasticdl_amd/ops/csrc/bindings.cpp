@@ -77,6 +77,16 @@ void edl_gemm256_variant_bf16(const void*, const void*, void*, int, int,
                               int, int, hipStream_t);
 void edl_fused_sgd_bf16(void*, float*, float*, const void*, int64_t, float,
                         float, bool, float, float, hipStream_t);
+void edl_bn_stats(const void*, int64_t, int64_t, float*, float*, hipStream_t);
+void edl_bn_apply(const void*, void*, int64_t, int64_t, const float*,
+                  const float*, const float*, const float*, bool,
+                  hipStream_t);
+void edl_bn_bwd_reduce(const void*, const void*, const void*, int64_t,
+                       int64_t, const float*, const float*, float*, float*,
+                       hipStream_t);
+void edl_bn_bwd_apply(const void*, const void*, const void*, void*, int64_t,
+                      int64_t, const float*, const float*, const float*,
+                      hipStream_t);
 void edl_fused_adamw_bf16(void*, float*, float*, float*, const void*, int64_t,
                           float, float, float, float, float, float, float,
                           hipStream_t);
@@ -414,6 +424,71 @@ torch::Tensor gemm256_bench(torch::Tensor a, torch::Tensor b, int64_t bars) {
   return c;
 }
 
+// ------------------------------ batch norm ------------------------------
+static void check_bn_xc(const torch::Tensor& x, const char* n) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
+              n, " must be bf16 cuda");
+  TORCH_CHECK(x.dim() == 2 && x.is_contiguous(), n, " must be [R,C] contig");
+  int64_t C = x.size(1);
+  TORCH_CHECK(C >= 8 && C % 8 == 0 && 256 % (C / 8) == 0,
+              "bn kernels need C in {8,16,...,2048} (pow2-ish); got ", C);
+}
+
+std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
+  check_bn_xc(x, "x");
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto sum = torch::zeros({x.size(1)}, opts);
+  auto sumsq = torch::zeros({x.size(1)}, opts);
+  edl_bn_stats(x.data_ptr(), x.size(0), x.size(1), sum.data_ptr<float>(),
+               sumsq.data_ptr<float>(), cur_stream());
+  return {sum, sumsq};
+}
+
+torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
+                       torch::Tensor rstd, c10::optional<torch::Tensor> gamma,
+                       c10::optional<torch::Tensor> beta, bool relu) {
+  check_bn_xc(x, "x");
+  auto y = torch::empty_like(x);
+  edl_bn_apply(x.data_ptr(), y.data_ptr(), x.size(0), x.size(1),
+               mean.data_ptr<float>(), rstd.data_ptr<float>(),
+               gamma.has_value() ? gamma->data_ptr<float>() : nullptr,
+               beta.has_value() ? beta->data_ptr<float>() : nullptr, relu,
+               cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> bn_bwd_reduce(torch::Tensor x, torch::Tensor dy,
+                                         c10::optional<torch::Tensor> relu_out,
+                                         torch::Tensor mean,
+                                         torch::Tensor rstd) {
+  check_bn_xc(x, "x");
+  check_bn_xc(dy, "dy");
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto s1 = torch::zeros({x.size(1)}, opts);
+  auto s2 = torch::zeros({x.size(1)}, opts);
+  edl_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(),
+                    relu_out.has_value() ? relu_out->data_ptr() : nullptr,
+                    x.size(0), x.size(1), mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), s1.data_ptr<float>(),
+                    s2.data_ptr<float>(), cur_stream());
+  return {s1, s2};
+}
+
+torch::Tensor bn_bwd_apply(torch::Tensor x, torch::Tensor dy,
+                           c10::optional<torch::Tensor> relu_out,
+                           torch::Tensor c1, torch::Tensor c2,
+                           torch::Tensor c3) {
+  check_bn_xc(x, "x");
+  check_bn_xc(dy, "dy");
+  auto dx = torch::empty_like(dy);
+  edl_bn_bwd_apply(x.data_ptr(), dy.data_ptr(),
+                   relu_out.has_value() ? relu_out->data_ptr() : nullptr,
+                   dx.data_ptr(), x.size(0), x.size(1),
+                   c1.data_ptr<float>(), c2.data_ptr<float>(),
+                   c3.data_ptr<float>(), cur_stream());
+  return dx;
+}
+
 // ---------------------- worker-side fused optimizers --------------------
 void fused_sgd_bf16(torch::Tensor p, torch::Tensor master, torch::Tensor vel,
                     torch::Tensor g, double lr, double mu, bool nesterov,
@@ -472,6 +547,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_rows", &scatter_rows);
   m.def("gemm_bias_act", &gemm_bias_act);
   m.def("gemm256_bench", &gemm256_bench);
+  m.def("bn_stats", &bn_stats);
+  m.def("bn_apply", &bn_apply);
+  m.def("bn_bwd_reduce", &bn_bwd_reduce);
+  m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("fused_sgd_bf16", &fused_sgd_bf16);
   m.def("fused_adamw_bf16", &fused_adamw_bf16);
 }

@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "elasticdl_amd/ops/csrc/ps_kernels.hip",
         "elasticdl_amd/ops/csrc/train_kernels.hip",
         "elasticdl_amd/ops/csrc/gemm_bf16.hip",
+        "elasticdl_amd/ops/csrc/bn_kernels.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
