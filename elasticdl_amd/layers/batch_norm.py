@@ -34,17 +34,18 @@ def _supported(x: torch.Tensor) -> bool:
 
 
 class _FusedBNFn(torch.autograd.Function):
+    """Operates on the FLAT [R, C] NHWC view; the module does the 4-D
+    (de)view OUTSIDE the Function so inplace consumers (ReLU(inplace))
+    see a normal autograd view, not a custom-Function output view."""
+
     @staticmethod
-    def forward(ctx, x, weight, bias, mean, rstd):
+    def forward(ctx, xf, weight, bias, mean, rstd):
         from elasticdl_amd.ops import require_native
 
         C = require_native()
-        xf = _flat_nhwc(x)
         y = C.bn_apply(xf, mean, rstd, weight.float(), bias.float(), False)
         ctx.save_for_backward(xf, mean, rstd, weight)
-        ctx.shape = x.shape
-        return y.reshape(x.shape[0], x.shape[2], x.shape[3],
-                         x.shape[1]).permute(0, 3, 1, 2)
+        return y
 
     @staticmethod
     def backward(ctx, dy):
@@ -52,9 +53,7 @@ class _FusedBNFn(torch.autograd.Function):
 
         C = require_native()
         xf, mean, rstd, weight = ctx.saved_tensors
-        dyf = _flat_nhwc(
-            dy.contiguous(memory_format=torch.channels_last)
-        ).contiguous()
+        dyf = dy.contiguous()
         s1, s2 = C.bn_bwd_reduce(xf, dyf, None, mean, rstd)
         r = xf.shape[0]
         g = weight.float()
@@ -64,11 +63,9 @@ class _FusedBNFn(torch.autograd.Function):
         b = a * rstd * s2 / r
         c = a * (mean * rstd * s2 - s1) / r
         dxf = C.bn_bwd_apply(xf, dyf, None, a, b, c)
-        n, ch, h, w = ctx.shape
-        dx = dxf.reshape(n, h, w, ch).permute(0, 3, 1, 2)
         dweight = s2.to(weight.dtype)
         dbias = s1.to(weight.dtype)
-        return dx, dweight, dbias, None, None
+        return dxf, dweight, dbias, None, None
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):
@@ -98,7 +95,9 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
                 if self.num_batches_tracked is not None:
                     self.num_batches_tracked += 1
 
-        return _FusedBNFn.apply(x, self.weight, self.bias, mean, rstd)
+        n, ch, h, w = x.shape
+        yf = _FusedBNFn.apply(xf, self.weight, self.bias, mean, rstd)
+        return yf.reshape(n, h, w, ch).permute(0, 3, 1, 2)
 
 
 def convert_to_fused_bn(module: nn.Module) -> nn.Module:
