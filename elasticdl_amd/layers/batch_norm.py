@@ -54,14 +54,10 @@ class _FusedBNFn(torch.autograd.Function):
         C = require_native()
         xf, mean, rstd, weight = ctx.saved_tensors
         dyf = dy.contiguous()
-        s1, s2 = C.bn_bwd_reduce(xf, dyf, None, mean, rstd)
-        r = xf.shape[0]
-        g = weight.float()
-        # dx = a*dy - b*x + c with per-channel coefficients:
-        #   a = g*rstd ; b = g*rstd^2*s2/R ; c = g*rstd*(mean*rstd*s2 - s1)/R
-        a = g * rstd
-        b = a * rstd * s2 / r
-        c = a * (mean * rstd * s2 - s1) / r
+        # the finalize kernel emits the dx coefficients alongside the
+        # per-channel grads — no Python-side per-channel math
+        s1, s2, a, b, c = C.bn_bwd_reduce(xf, dyf, None, mean, rstd,
+                                          weight.float())
         dxf = C.bn_bwd_apply(xf, dyf, None, a, b, c)
         dweight = s2.to(weight.dtype)
         dbias = s1.to(weight.dtype)
@@ -81,10 +77,7 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
         C = require_native()
         xf = _flat_nhwc(x)
         r = xf.shape[0]
-        sum_, sumsq = C.bn_stats(xf)
-        mean = sum_ / r
-        var = (sumsq / r - mean * mean).clamp_(min=0)
-        rstd = (var + self.eps).rsqrt()
+        mean, var, rstd = C.bn_stats(xf, self.eps)
 
         if self.track_running_stats and self.running_mean is not None:
             with torch.no_grad():

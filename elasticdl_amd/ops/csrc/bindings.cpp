@@ -77,12 +77,15 @@ void edl_gemm256_variant_bf16(const void*, const void*, void*, int, int,
                               int, int, hipStream_t);
 void edl_fused_sgd_bf16(void*, float*, float*, const void*, int64_t, float,
                         float, bool, float, float, hipStream_t);
-void edl_bn_stats(const void*, int64_t, int64_t, float*, float*, hipStream_t);
+int edl_bn_grid_for(int64_t, int64_t);
+void edl_bn_stats(const void*, int64_t, int64_t, float*, int, float, float*,
+                  float*, float*, hipStream_t);
 void edl_bn_apply(const void*, void*, int64_t, int64_t, const float*,
                   const float*, const float*, const float*, bool,
                   hipStream_t);
 void edl_bn_bwd_reduce(const void*, const void*, const void*, int64_t,
-                       int64_t, const float*, const float*, float*, float*,
+                       int64_t, const float*, const float*, const float*,
+                       float*, int, float*, float*, float*, float*, float*,
                        hipStream_t);
 void edl_bn_bwd_apply(const void*, const void*, const void*, void*, int64_t,
                       int64_t, const float*, const float*, const float*,
@@ -434,14 +437,20 @@ static void check_bn_xc(const torch::Tensor& x, const char* n) {
               "bn kernels need C in {8,16,...,2048} (pow2-ish); got ", C);
 }
 
-std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
+// returns (mean, var_biased, rstd)
+std::vector<torch::Tensor> bn_stats(torch::Tensor x, double eps) {
   check_bn_xc(x, "x");
   auto opts = x.options().dtype(torch::kFloat32);
-  auto sum = torch::zeros({x.size(1)}, opts);
-  auto sumsq = torch::zeros({x.size(1)}, opts);
-  edl_bn_stats(x.data_ptr(), x.size(0), x.size(1), sum.data_ptr<float>(),
-               sumsq.data_ptr<float>(), cur_stream());
-  return {sum, sumsq};
+  int64_t C = x.size(1);
+  int G = edl_bn_grid_for(x.size(0), C);
+  auto part = torch::empty({(int64_t)G, 2 * C}, opts);
+  auto mean = torch::empty({C}, opts);
+  auto var = torch::empty({C}, opts);
+  auto rstd = torch::empty({C}, opts);
+  edl_bn_stats(x.data_ptr(), x.size(0), C, part.data_ptr<float>(), G,
+               static_cast<float>(eps), mean.data_ptr<float>(),
+               var.data_ptr<float>(), rstd.data_ptr<float>(), cur_stream());
+  return {mean, var, rstd};
 }
 
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
@@ -457,21 +466,34 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
   return y;
 }
 
+// returns (sum_dy=dbeta, sum_dy_xhat=dgamma, a, b, c) — a/b/c are the
+// dx = a*dy - b*x + c per-channel coefficients
 std::vector<torch::Tensor> bn_bwd_reduce(torch::Tensor x, torch::Tensor dy,
                                          c10::optional<torch::Tensor> relu_out,
                                          torch::Tensor mean,
-                                         torch::Tensor rstd) {
+                                         torch::Tensor rstd,
+                                         c10::optional<torch::Tensor> gamma) {
   check_bn_xc(x, "x");
   check_bn_xc(dy, "dy");
   auto opts = x.options().dtype(torch::kFloat32);
-  auto s1 = torch::zeros({x.size(1)}, opts);
-  auto s2 = torch::zeros({x.size(1)}, opts);
+  int64_t C = x.size(1);
+  int G = edl_bn_grid_for(x.size(0), C);
+  auto part = torch::empty({(int64_t)G, 2 * C}, opts);
+  auto s1 = torch::empty({C}, opts);
+  auto s2 = torch::empty({C}, opts);
+  auto ca = torch::empty({C}, opts);
+  auto cb = torch::empty({C}, opts);
+  auto cc = torch::empty({C}, opts);
   edl_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(),
                     relu_out.has_value() ? relu_out->data_ptr() : nullptr,
-                    x.size(0), x.size(1), mean.data_ptr<float>(),
-                    rstd.data_ptr<float>(), s1.data_ptr<float>(),
-                    s2.data_ptr<float>(), cur_stream());
-  return {s1, s2};
+                    x.size(0), C, mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(),
+                    gamma.has_value() ? gamma->data_ptr<float>() : nullptr,
+                    part.data_ptr<float>(), G, s1.data_ptr<float>(),
+                    s2.data_ptr<float>(), ca.data_ptr<float>(),
+                    cb.data_ptr<float>(), cc.data_ptr<float>(),
+                    cur_stream());
+  return {s1, s2, ca, cb, cc};
 }
 
 torch::Tensor bn_bwd_apply(torch::Tensor x, torch::Tensor dy,

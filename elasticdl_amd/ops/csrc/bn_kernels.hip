@@ -35,10 +35,13 @@ static inline int bn_grid(int64_t rows_per_iter_total) {
 }
 
 // ---------------------------------------------------------------- stats
-// sum[C], sumsq[C] must be zeroed by the caller.
+// Two-stage: blocks write per-block partial sums to part[G][2C] (plain
+// stores — a single atomic target per channel serializes ~2048 blocks
+// and put a ~470 us floor under every layer), then a finalize kernel
+// reduces partials and emits mean/rstd/var directly.
 __global__ __launch_bounds__(BN_THREADS) void bn_stats_kernel(
     const __bf16* __restrict__ x, int64_t R, int64_t C,
-    float* __restrict__ sum, float* __restrict__ sumsq) {
+    float* __restrict__ part) {
   const int cpt = (int)(C >> 3);           // threads per row
   const int rpi = BN_THREADS / cpt;        // rows per block-iter
   const int cg = threadIdx.x % cpt;        // channel group
@@ -68,7 +71,8 @@ __global__ __launch_bounds__(BN_THREADS) void bn_stats_kernel(
       for (int k = 0; k < 8; ++k)
         s[k] += red[(rr * cpt + cg) * 8 + k];
 #pragma unroll
-    for (int k = 0; k < 8; ++k) atomicAdd(&sum[c0 + k], s[k]);
+    for (int k = 0; k < 8; ++k)
+      part[(int64_t)blockIdx.x * 2 * C + c0 + k] = s[k];
   }
   __syncthreads();
 #pragma unroll
@@ -80,7 +84,41 @@ __global__ __launch_bounds__(BN_THREADS) void bn_stats_kernel(
       for (int k = 0; k < 8; ++k)
         q[k] += red[(rr * cpt + cg) * 8 + k];
 #pragma unroll
-    for (int k = 0; k < 8; ++k) atomicAdd(&sumsq[c0 + k], q[k]);
+    for (int k = 0; k < 8; ++k)
+      part[(int64_t)blockIdx.x * 2 * C + C + c0 + k] = q[k];
+  }
+}
+
+// finalize: one block per channel; threads parallel-reduce the G partial
+// rows, lane 0 writes mean / biased var / rstd.
+__global__ __launch_bounds__(BN_THREADS) void bn_stats_finalize_kernel(
+    const float* __restrict__ part, int G, int64_t C, int64_t R, float eps,
+    float* __restrict__ mean, float* __restrict__ var,
+    float* __restrict__ rstd) {
+  const int64_t c = blockIdx.x;
+  float s = 0.f, q = 0.f;
+  for (int g = threadIdx.x; g < G; g += BN_THREADS) {
+    s += part[(int64_t)g * 2 * C + c];
+    q += part[(int64_t)g * 2 * C + C + c];
+  }
+  __shared__ float red[2 * BN_THREADS];
+  red[threadIdx.x] = s;
+  red[BN_THREADS + threadIdx.x] = q;
+  __syncthreads();
+  for (int off = BN_THREADS / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      red[threadIdx.x] += red[threadIdx.x + off];
+      red[BN_THREADS + threadIdx.x] += red[BN_THREADS + threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    float m = red[0] / (float)R;
+    float v = red[BN_THREADS] / (float)R - m * m;
+    if (v < 0.f) v = 0.f;
+    mean[c] = m;
+    var[c] = v;
+    rstd[c] = rsqrtf(v + eps);
   }
 }
 
@@ -126,7 +164,7 @@ __global__ __launch_bounds__(BN_THREADS) void bn_bwd_reduce_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ dy,
     const __bf16* __restrict__ relu_out, int64_t R, int64_t C,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ sum_dy, float* __restrict__ sum_dy_xhat) {
+    float* __restrict__ part) {
   const int cpt = (int)(C >> 3);
   const int rpi = BN_THREADS / cpt;
   const int cg = threadIdx.x % cpt;
@@ -166,7 +204,8 @@ __global__ __launch_bounds__(BN_THREADS) void bn_bwd_reduce_kernel(
       for (int k = 0; k < 8; ++k)
         s1[k] += red[(rr * cpt + cg) * 8 + k];
 #pragma unroll
-    for (int k = 0; k < 8; ++k) atomicAdd(&sum_dy[c0 + k], s1[k]);
+    for (int k = 0; k < 8; ++k)
+      part[(int64_t)blockIdx.x * 2 * C + c0 + k] = s1[k];
   }
   __syncthreads();
 #pragma unroll
@@ -178,7 +217,49 @@ __global__ __launch_bounds__(BN_THREADS) void bn_bwd_reduce_kernel(
       for (int k = 0; k < 8; ++k)
         s2[k] += red[(rr * cpt + cg) * 8 + k];
 #pragma unroll
-    for (int k = 0; k < 8; ++k) atomicAdd(&sum_dy_xhat[c0 + k], s2[k]);
+    for (int k = 0; k < 8; ++k)
+      part[(int64_t)blockIdx.x * 2 * C + C + c0 + k] = s2[k];
+  }
+}
+
+// finalize: reduce partials and emit both the per-channel grads
+// (dgamma = sum_dy_xhat, dbeta = sum_dy) and the dx coefficients
+//   a = g*rstd ; b = a*rstd*s2/R ; c = a*(mean*rstd*s2 - s1)/R
+// so the backward needs no Python-side per-channel math.
+__global__ __launch_bounds__(BN_THREADS) void bn_bwd_finalize_kernel(
+    const float* __restrict__ part, int G, int64_t C, int64_t R,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ gamma, float* __restrict__ sum_dy,
+    float* __restrict__ sum_dy_xhat, float* __restrict__ ca,
+    float* __restrict__ cb, float* __restrict__ cc) {
+  const int64_t c = blockIdx.x;
+  float s1 = 0.f, s2 = 0.f;
+  for (int g = threadIdx.x; g < G; g += BN_THREADS) {
+    s1 += part[(int64_t)g * 2 * C + c];
+    s2 += part[(int64_t)g * 2 * C + C + c];
+  }
+  __shared__ float red[2 * BN_THREADS];
+  red[threadIdx.x] = s1;
+  red[BN_THREADS + threadIdx.x] = s2;
+  __syncthreads();
+  for (int off = BN_THREADS / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      red[threadIdx.x] += red[threadIdx.x + off];
+      red[BN_THREADS + threadIdx.x] += red[BN_THREADS + threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    s1 = red[0];
+    s2 = red[BN_THREADS];
+    sum_dy[c] = s1;
+    sum_dy_xhat[c] = s2;
+    float g = gamma ? gamma[c] : 1.f;
+    float rs = rstd[c];
+    float a = g * rs;
+    ca[c] = a;
+    cb[c] = a * rs * s2 / (float)R;
+    cc[c] = a * (mean[c] * rs * s2 - s1) / (float)R;
   }
 }
 
@@ -224,12 +305,18 @@ __global__ __launch_bounds__(BN_THREADS) void bn_bwd_apply_kernel(
 
 extern "C" {
 
-void edl_bn_stats(const void* x, int64_t R, int64_t C, float* sum,
-                  float* sumsq, hipStream_t s) {
-  int cpt = (int)(C >> 3);
-  int rpi = BN_THREADS / cpt;
-  bn_stats_kernel<<<bn_grid((R + rpi - 1) / rpi), BN_THREADS, 0, s>>>(
-      reinterpret_cast<const __bf16*>(x), R, C, sum, sumsq);
+int edl_bn_grid_for(int64_t R, int64_t C) {
+  int rpi = BN_THREADS / (int)(C >> 3);
+  return bn_grid((R + rpi - 1) / rpi);
+}
+
+void edl_bn_stats(const void* x, int64_t R, int64_t C, float* part, int G,
+                  float eps, float* mean, float* var, float* rstd,
+                  hipStream_t s) {
+  bn_stats_kernel<<<G, BN_THREADS, 0, s>>>(
+      reinterpret_cast<const __bf16*>(x), R, C, part);
+  bn_stats_finalize_kernel<<<(int)C, BN_THREADS, 0, s>>>(part, G, C, R, eps,
+                                                         mean, var, rstd);
 }
 
 void edl_bn_apply(const void* x, void* y, int64_t R, int64_t C,
@@ -244,14 +331,14 @@ void edl_bn_apply(const void* x, void* y, int64_t R, int64_t C,
 
 void edl_bn_bwd_reduce(const void* x, const void* dy, const void* relu_out,
                        int64_t R, int64_t C, const float* mean,
-                       const float* rstd, float* sum_dy, float* sum_dy_xhat,
-                       hipStream_t s) {
-  int cpt = (int)(C >> 3);
-  int rpi = BN_THREADS / cpt;
-  bn_bwd_reduce_kernel<<<bn_grid((R + rpi - 1) / rpi), BN_THREADS, 0, s>>>(
+                       const float* rstd, const float* gamma, float* part,
+                       int G, float* sum_dy, float* sum_dy_xhat, float* ca,
+                       float* cb, float* cc, hipStream_t s) {
+  bn_bwd_reduce_kernel<<<G, BN_THREADS, 0, s>>>(
       reinterpret_cast<const __bf16*>(x), reinterpret_cast<const __bf16*>(dy),
-      reinterpret_cast<const __bf16*>(relu_out), R, C, mean, rstd, sum_dy,
-      sum_dy_xhat);
+      reinterpret_cast<const __bf16*>(relu_out), R, C, mean, rstd, part);
+  bn_bwd_finalize_kernel<<<(int)C, BN_THREADS, 0, s>>>(
+      part, G, C, R, mean, rstd, gamma, sum_dy, sum_dy_xhat, ca, cb, cc);
 }
 
 void edl_bn_bwd_apply(const void* x, const void* dy, const void* relu_out,

@@ -58,13 +58,13 @@ def test_bn_kernels_match_torch_oracle(C):
     torch.manual_seed(0)
     R = 1024
     x = torch.randn(R, C, device="cuda", dtype=torch.bfloat16)
-    s, q = K.bn_stats(x)
+    mean, var, rstd = K.bn_stats(x, 1e-5)
     xf = x.float()
-    assert torch.allclose(s, xf.sum(0), rtol=1e-3, atol=1e-2)
-    assert torch.allclose(q, (xf * xf).sum(0), rtol=1e-3, atol=1e-1)
-
-    mean = s / R
-    rstd = (q / R - mean * mean + 1e-5).rsqrt()
+    assert torch.allclose(mean, xf.mean(0), rtol=1e-3, atol=1e-3)
+    assert torch.allclose(var, xf.var(0, unbiased=False), rtol=1e-2,
+                          atol=1e-3)
+    assert torch.allclose(rstd, (xf.var(0, unbiased=False) + 1e-5).rsqrt(),
+                          rtol=1e-2, atol=1e-3)
     gamma = torch.randn(C, device="cuda")
     beta = torch.randn(C, device="cuda")
     y = K.bn_apply(x, mean, rstd, gamma, beta, False)
@@ -75,11 +75,17 @@ def test_bn_kernels_match_torch_oracle(C):
     assert torch.allclose(yr.float(), ref.clamp(min=0), atol=0.05, rtol=0.05)
 
     dy = torch.randn_like(x)
-    s1, s2 = K.bn_bwd_reduce(x, dy, None, mean, rstd)
+    s1, s2, a, b, c = K.bn_bwd_reduce(x, dy, None, mean, rstd, gamma)
     dyf = dy.float()
     xhat = (xf - mean) * rstd
     assert torch.allclose(s1, dyf.sum(0), rtol=1e-3, atol=1e-1)
     assert torch.allclose(s2, (dyf * xhat).sum(0), rtol=1e-2, atol=2e-1)
+    assert torch.allclose(a, gamma * rstd, rtol=1e-4, atol=1e-5)
+    dx = K.bn_bwd_apply(x, dy, None, a, b, c)
+    ref_dx = gamma * rstd * (dyf - s1 / R - xhat * s2 / R)
+    assert torch.allclose(dx.float(), ref_dx, atol=0.08, rtol=0.05), (
+        (dx.float() - ref_dx).abs().max()
+    )
 
 
 @pytest.mark.gpu
