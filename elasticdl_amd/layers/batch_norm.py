@@ -35,16 +35,23 @@ def _supported(x: torch.Tensor) -> bool:
 
 class _FusedBNFn(torch.autograd.Function):
     """Operates on the FLAT [R, C] NHWC view; the module does the 4-D
-    (de)view OUTSIDE the Function so inplace consumers (ReLU(inplace))
-    see a normal autograd view, not a custom-Function output view."""
+    (de)view OUTSIDE the Function so inplace consumers see a normal
+    autograd view, not a custom-Function output view. With ``relu`` the
+    activation is fused into the normalize kernel and its backward mask
+    into the reduce/apply kernels (one less read+write of the activation
+    in each direction)."""
 
     @staticmethod
-    def forward(ctx, xf, weight, bias, mean, rstd):
+    def forward(ctx, xf, weight, bias, mean, rstd, relu):
         from elasticdl_amd.ops import require_native
 
         C = require_native()
-        y = C.bn_apply(xf, mean, rstd, weight.float(), bias.float(), False)
-        ctx.save_for_backward(xf, mean, rstd, weight)
+        y = C.bn_apply(xf, mean, rstd, weight.float(), bias.float(), relu)
+        if relu:
+            ctx.save_for_backward(xf, mean, rstd, weight, y)
+        else:
+            ctx.save_for_backward(xf, mean, rstd, weight)
+        ctx.relu = relu
         return y
 
     @staticmethod
@@ -52,25 +59,40 @@ class _FusedBNFn(torch.autograd.Function):
         from elasticdl_amd.ops import require_native
 
         C = require_native()
-        xf, mean, rstd, weight = ctx.saved_tensors
+        if ctx.relu:
+            xf, mean, rstd, weight, y = ctx.saved_tensors
+        else:
+            xf, mean, rstd, weight = ctx.saved_tensors
+            y = None
         dyf = dy.contiguous()
         # the finalize kernel emits the dx coefficients alongside the
         # per-channel grads — no Python-side per-channel math
-        s1, s2, a, b, c = C.bn_bwd_reduce(xf, dyf, None, mean, rstd,
+        s1, s2, a, b, c = C.bn_bwd_reduce(xf, dyf, y, mean, rstd,
                                           weight.float())
-        dxf = C.bn_bwd_apply(xf, dyf, None, a, b, c)
+        dxf = C.bn_bwd_apply(xf, dyf, y, a, b, c)
         dweight = s2.to(weight.dtype)
         dbias = s1.to(weight.dtype)
-        return dxf, dweight, dbias, None, None
+        return dxf, dweight, dbias, None, None, None
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):
+    """Set ``fuse_relu=True`` (or use the BNReLU alias) to fold the
+    following ReLU into the BN kernels — fallback paths apply
+    F.relu after stock BN so module semantics are identical."""
+
+    fuse_relu = False
+
+    def __init__(self, *args, fuse_relu: bool = False, **kw):
+        super().__init__(*args, **kw)
+        self.fuse_relu = fuse_relu
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         from elasticdl_amd.ops import use_native
 
         if not (self.training and self.affine and _supported(x)
                 and use_native(x.device)):
-            return super().forward(x)
+            y = super().forward(x)
+            return torch.relu_(y) if self.fuse_relu else y
 
         from elasticdl_amd.ops import require_native
 
@@ -89,8 +111,15 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
                     self.num_batches_tracked += 1
 
         n, ch, h, w = x.shape
-        yf = _FusedBNFn.apply(xf, self.weight, self.bias, mean, rstd)
+        yf = _FusedBNFn.apply(xf, self.weight, self.bias, mean, rstd,
+                              self.fuse_relu)
         return yf.reshape(n, h, w, ch).permute(0, 3, 1, 2)
+
+
+def BNReLU(num_features: int, **kw) -> FusedBatchNorm2d:
+    """BatchNorm2d + ReLU as one fused module (state_dict-compatible
+    with a plain BatchNorm2d of the same name)."""
+    return FusedBatchNorm2d(num_features, fuse_relu=True, **kw)
 
 
 def convert_to_fused_bn(module: nn.Module) -> nn.Module:

@@ -19,20 +19,22 @@ class Bottleneck(nn.Module):
     def __init__(self, in_ch: int, width: int, stride: int = 1,
                  downsample: Optional[nn.Module] = None):
         super().__init__()
+        from elasticdl_amd.layers.batch_norm import BNReLU, FusedBatchNorm2d
+
         out_ch = width * self.expansion
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = BNReLU(width)  # BN+ReLU fused in one kernel pass
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = BNReLU(width)
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(out_ch)
+        self.bn3 = FusedBatchNorm2d(out_ch)  # relu comes after the residual
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
 
     def forward(self, x):
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
         out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
@@ -43,9 +45,10 @@ class ResNet(nn.Module):
     def __init__(self, layers: List[int], num_classes: int = 1000):
         super().__init__()
         self.in_ch = 64
+        from elasticdl_amd.layers.batch_norm import BNReLU
+
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = BNReLU(64)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, layers[0], 1)
         self.layer2 = self._make_layer(128, layers[1], 2)
@@ -65,9 +68,11 @@ class ResNet(nn.Module):
         downsample = None
         out_ch = width * Bottleneck.expansion
         if stride != 1 or self.in_ch != out_ch:
+            from elasticdl_amd.layers.batch_norm import FusedBatchNorm2d
+
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(out_ch),
+                FusedBatchNorm2d(out_ch),
             )
         layers = [Bottleneck(self.in_ch, width, stride, downsample)]
         self.in_ch = out_ch
@@ -76,22 +81,18 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)
 
 
-def resnet50(num_classes: int = 1000, fused_bn: bool = True) -> ResNet:
-    model = ResNet([3, 4, 6, 3], num_classes)
-    if fused_bn:
-        # MI355X-native channels_last BN kernels (torch's are the
-        # dominant step cost — profiles/resnet_r02.md); a transparent
-        # fallback keeps CPU/fp32 paths on stock BN
-        from elasticdl_amd.layers.batch_norm import convert_to_fused_bn
-
-        convert_to_fused_bn(model)
-    return model
+def resnet50(num_classes: int = 1000) -> ResNet:
+    """ResNet-50 built on the MI355X-native fused BN(+ReLU) modules
+    (torch's channels_last BN kernels were the dominant step cost —
+    profiles/resnet_r02.md); the modules fall back to stock BN on
+    CPU/fp32, so the same model runs everywhere."""
+    return ResNet([3, 4, 6, 3], num_classes)
 
 
 def resnet18_cifar(num_classes: int = 10) -> ResNet:

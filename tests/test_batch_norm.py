@@ -140,3 +140,36 @@ def test_resnet_block_trains_with_fused_bn_gpu():
     assert torch.isfinite(loss)
     grads = [p.grad for p in m.parameters() if p.grad is not None]
     assert all(torch.isfinite(g).all() for g in grads)
+
+
+@pytest.mark.gpu
+def test_bnrelu_module_matches_fp32_oracle_gpu():
+    from elasticdl_amd.layers.batch_norm import BNReLU
+
+    torch.manual_seed(3)
+    C = 64
+    ref = nn.BatchNorm2d(C).cuda()
+    fused = BNReLU(C).cuda()
+    fused.load_state_dict(ref.state_dict())
+    fused = fused.to(torch.bfloat16)
+
+    x32 = torch.randn(8, C, 14, 14, device="cuda", requires_grad=True)
+    xbf = x32.detach().to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last
+    ).requires_grad_(True)
+
+    y_ref = torch.relu(ref(x32))
+    y = fused(xbf)
+    assert torch.allclose(y.float(), y_ref, atol=0.1, rtol=0.05)
+    assert (y.float() >= 0).all()
+
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y.backward(g.to(torch.bfloat16))
+    assert torch.allclose(xbf.grad.float(), x32.grad, atol=0.1, rtol=0.1), (
+        (xbf.grad.float() - x32.grad).abs().max()
+    )
+    assert torch.allclose(fused.weight.grad.float(), ref.weight.grad,
+                          atol=0.5, rtol=0.05)
+    assert torch.allclose(fused.bias.grad.float(), ref.bias.grad,
+                          atol=0.5, rtol=0.05)
