@@ -158,7 +158,8 @@ def test_bnrelu_module_matches_fp32_oracle_gpu():
         memory_format=torch.channels_last
     ).requires_grad_(True)
 
-    y_ref = torch.relu(ref(x32))
+    pre = ref(x32)
+    y_ref = torch.relu(pre)
     y = fused(xbf)
     assert torch.allclose(y.float(), y_ref, atol=0.1, rtol=0.05)
     assert (y.float() >= 0).all()
@@ -166,9 +167,11 @@ def test_bnrelu_module_matches_fp32_oracle_gpu():
     g = torch.randn_like(y_ref)
     y_ref.backward(g)
     y.backward(g.to(torch.bfloat16))
-    assert torch.allclose(xbf.grad.float(), x32.grad, atol=0.1, rtol=0.1), (
-        (xbf.grad.float() - x32.grad).abs().max()
-    )
+    # at |pre-activation| ~ bf16 epsilon the ReLU mask legitimately
+    # disagrees between bf16 and fp32 — compare away from the boundary
+    interior = (pre.detach().abs() > 0.02)
+    diff = (xbf.grad.float() - x32.grad).abs() * interior
+    assert diff.max() < 0.1, diff.max()
     assert torch.allclose(fused.weight.grad.float(), ref.weight.grad,
                           atol=0.5, rtol=0.05)
     assert torch.allclose(fused.bias.grad.float(), ref.bias.grad,
