@@ -116,6 +116,46 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
         return yf.reshape(n, h, w, ch).permute(0, 3, 1, 2)
 
 
+class _AddReluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        from elasticdl_amd.ops import require_native
+
+        C = require_native()
+        z = C.add_relu_fwd(a, b)
+        ctx.save_for_backward(z)
+        return z
+
+    @staticmethod
+    def backward(ctx, dz):
+        from elasticdl_amd.ops import require_native
+
+        C = require_native()
+        (z,) = ctx.saved_tensors
+        if dz.stride() != z.stride():
+            dz = dz.contiguous(memory_format=torch.channels_last) \
+                if z.is_contiguous(memory_format=torch.channels_last) \
+                else dz.contiguous()
+        dg = C.add_relu_bwd(dz, z)
+        return dg, dg
+
+
+def add_relu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """relu(a + b) as ONE kernel each direction (the residual join of
+    every ResNet block; torch launches separate add / relu /
+    threshold_backward passes). Falls back to torch when the fused path
+    doesn't apply (CPU, fp32, mismatched layouts)."""
+    from elasticdl_amd.ops import use_native
+
+    if (
+        a.is_cuda and a.dtype == torch.bfloat16 and b.dtype == a.dtype
+        and a.stride() == b.stride() and a.numel() % 8 == 0
+        and use_native(a.device)
+    ):
+        return _AddReluFn.apply(a, b)
+    return torch.relu_(a + b)
+
+
 def BNReLU(num_features: int, **kw) -> FusedBatchNorm2d:
     """BatchNorm2d + ReLU as one fused module (state_dict-compatible
     with a plain BatchNorm2d of the same name)."""

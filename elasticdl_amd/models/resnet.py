@@ -32,13 +32,15 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
+        from elasticdl_amd.layers.batch_norm import add_relu
+
         identity = x
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        return self.relu(out + identity)
+        return add_relu(out, identity)
 
 
 class ResNet(nn.Module):

@@ -78,6 +78,8 @@ void edl_gemm256_variant_bf16(const void*, const void*, void*, int, int,
 void edl_fused_sgd_bf16(void*, float*, float*, const void*, int64_t, float,
                         float, bool, float, float, hipStream_t);
 int edl_bn_grid_for(int64_t, int64_t);
+void edl_add_relu_fwd(const void*, const void*, void*, int64_t, hipStream_t);
+void edl_add_relu_bwd(const void*, const void*, void*, int64_t, hipStream_t);
 void edl_bn_stats(const void*, int64_t, int64_t, float*, int, float, float*,
                   float*, float*, hipStream_t);
 void edl_bn_apply(const void*, void*, int64_t, int64_t, const float*,
@@ -437,6 +439,23 @@ static void check_bn_xc(const torch::Tensor& x, const char* n) {
               "bn kernels need C in {8,16,...,2048} (pow2-ish); got ", C);
 }
 
+torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() % 8 == 0);
+  auto z = torch::empty_like(a);
+  edl_add_relu_fwd(a.data_ptr(), b.data_ptr(), z.data_ptr(), a.numel(),
+                   cur_stream());
+  return z;
+}
+
+torch::Tensor add_relu_bwd(torch::Tensor dz, torch::Tensor z) {
+  TORCH_CHECK(dz.numel() == z.numel() && dz.numel() % 8 == 0);
+  auto dg = torch::empty_like(dz);
+  edl_add_relu_bwd(dz.data_ptr(), z.data_ptr(), dg.data_ptr(), dz.numel(),
+                   cur_stream());
+  return dg;
+}
+
 // returns (mean, var_biased, rstd)
 std::vector<torch::Tensor> bn_stats(torch::Tensor x, double eps) {
   check_bn_xc(x, "x");
@@ -570,6 +589,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bias_act", &gemm_bias_act);
   m.def("gemm256_bench", &gemm256_bench);
   m.def("bn_stats", &bn_stats);
+  m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("add_relu_bwd", &add_relu_bwd);
   m.def("bn_apply", &bn_apply);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_apply", &bn_bwd_apply);

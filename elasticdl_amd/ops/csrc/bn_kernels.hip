@@ -303,7 +303,63 @@ __global__ __launch_bounds__(BN_THREADS) void bn_bwd_apply_kernel(
   }
 }
 
+// ------------------------------------------------------- add + relu
+// z = relu(a + b): the residual-join of every ResNet block, fused into
+// one pass (torch runs add and relu as separate activation-sized
+// kernels). Backward: da = db = dz * (z > 0), also one pass.
+__global__ __launch_bounds__(BN_THREADS) void add_relu_fwd_kernel(
+    const __bf16* __restrict__ a, const __bf16* __restrict__ b,
+    __bf16* __restrict__ z, int64_t n8) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    bn_bf16x8 va = reinterpret_cast<const bn_bf16x8*>(a)[i];
+    bn_bf16x8 vb = reinterpret_cast<const bn_bf16x8*>(b)[i];
+    bn_bf16x8 o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float f = (float)va[k] + (float)vb[k];
+      o[k] = (__bf16)(f > 0.f ? f : 0.f);
+    }
+    reinterpret_cast<bn_bf16x8*>(z)[i] = o;
+  }
+}
+
+__global__ __launch_bounds__(BN_THREADS) void add_relu_bwd_kernel(
+    const __bf16* __restrict__ dz, const __bf16* __restrict__ z,
+    __bf16* __restrict__ dg, int64_t n8) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    bn_bf16x8 vd = reinterpret_cast<const bn_bf16x8*>(dz)[i];
+    bn_bf16x8 vz = reinterpret_cast<const bn_bf16x8*>(z)[i];
+    bn_bf16x8 o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k)
+      o[k] = (float)vz[k] > 0.f ? vd[k] : (__bf16)0.f;
+    reinterpret_cast<bn_bf16x8*>(dg)[i] = o;
+  }
+}
+
 extern "C" {
+
+void edl_add_relu_fwd(const void* a, const void* b, void* z, int64_t n,
+                      hipStream_t s) {
+  int64_t n8 = n >> 3;
+  add_relu_fwd_kernel<<<bn_grid((n8 + BN_THREADS - 1) / BN_THREADS),
+                        BN_THREADS, 0, s>>>(
+      reinterpret_cast<const __bf16*>(a), reinterpret_cast<const __bf16*>(b),
+      reinterpret_cast<__bf16*>(z), n8);
+}
+
+void edl_add_relu_bwd(const void* dz, const void* z, void* dg, int64_t n,
+                      hipStream_t s) {
+  int64_t n8 = n >> 3;
+  add_relu_bwd_kernel<<<bn_grid((n8 + BN_THREADS - 1) / BN_THREADS),
+                        BN_THREADS, 0, s>>>(
+      reinterpret_cast<const __bf16*>(dz), reinterpret_cast<const __bf16*>(z),
+      reinterpret_cast<__bf16*>(dg), n8);
+}
 
 int edl_bn_grid_for(int64_t R, int64_t C) {
   int rpi = BN_THREADS / (int)(C >> 3);

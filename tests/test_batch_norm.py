@@ -176,3 +176,33 @@ def test_bnrelu_module_matches_fp32_oracle_gpu():
                           atol=0.5, rtol=0.05)
     assert torch.allclose(fused.bias.grad.float(), ref.bias.grad,
                           atol=0.5, rtol=0.05)
+
+
+def test_add_relu_cpu_fallback():
+    from elasticdl_amd.layers.batch_norm import add_relu
+
+    a = torch.randn(4, 8, requires_grad=True)
+    b = torch.randn(4, 8, requires_grad=True)
+    z = add_relu(a.clone(), b)  # clone: fallback relu_ is inplace on a+b
+    assert torch.allclose(z, torch.relu(a + b))
+
+
+@pytest.mark.gpu
+def test_add_relu_gpu_matches_oracle():
+    from elasticdl_amd.layers.batch_norm import add_relu
+
+    torch.manual_seed(5)
+    a32 = torch.randn(64, 128, device="cuda", requires_grad=True)
+    b32 = torch.randn(64, 128, device="cuda", requires_grad=True)
+    a = a32.detach().to(torch.bfloat16).requires_grad_(True)
+    b = b32.detach().to(torch.bfloat16).requires_grad_(True)
+    z = add_relu(a, b)
+    z_ref = torch.relu(a32 + b32)
+    assert torch.allclose(z.float(), z_ref, atol=0.05, rtol=0.05)
+    g = torch.randn_like(z_ref)
+    z_ref.backward(g)
+    z.backward(g.to(torch.bfloat16))
+    pre = (a32 + b32).detach()
+    interior = pre.abs() > 0.02  # bf16 relu-mask boundary
+    assert ((a.grad.float() - a32.grad).abs() * interior).max() < 0.05
+    assert ((b.grad.float() - b32.grad).abs() * interior).max() < 0.05
