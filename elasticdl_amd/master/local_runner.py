@@ -115,7 +115,29 @@ class LocalProcessManager:
             cmd = self.ps_command(i)
             self._spawn(f"ps-{i}", cmd, "ps", i, {})
 
+    def _wait_ps_ready(self, timeout: float = 60.0) -> None:
+        """Block until every PS port accepts connections. Without this,
+        the first workers race the PS processes' startup and burn their
+        minibatch-retry budget on connection-refused RPCs (in k8s the
+        PS Services play this role; locally we gate explicitly)."""
+        import socket
+        import time as _time
+
+        deadline = _time.monotonic() + timeout
+        for addr in self.ps_addrs:
+            host, _, port = addr.rpartition(":")
+            while _time.monotonic() < deadline:
+                try:
+                    with socket.create_connection((host, int(port)), 1.0):
+                        break
+                except OSError:
+                    _time.sleep(0.2)
+            else:
+                logger.warning("PS %s not ready after %.0fs", addr, timeout)
+
     def start_workers(self) -> None:
+        if self.num_ps and self.ps_addrs:
+            self._wait_ps_ready()
         for _ in range(self.num_workers):
             self.start_one_worker()
 

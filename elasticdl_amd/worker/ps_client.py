@@ -28,6 +28,20 @@ class PSClient:
         self._clients = [RpcClient(a) for a in ps_addrs]
         # dense param name -> shard cache
         self._name_shard: Dict[str, int] = {}
+        self._ready = False
+
+    def _ensure_ready(self, timeout: float = 60.0) -> None:
+        """Wait for every PS channel once before the first RPC — a worker
+        that starts (or is relaunched) while PS pods are still coming up
+        must not burn its retry budget on connection-refused errors."""
+        if self._ready:
+            return
+        for c in self._clients:
+            try:
+                c.wait_ready(timeout)
+            except Exception:  # noqa: BLE001 - let the RPC surface the error
+                pass
+        self._ready = True
 
     def _shard_of_name(self, name: str) -> int:
         s = self._name_shard.get(name)
@@ -39,6 +53,7 @@ class PSClient:
     # ------------------------------------------------------------- model init
     def push_model(self, dense: Dict[str, torch.Tensor],
                    embedding_infos: List[dict]) -> None:
+        self._ensure_ready()
         per_shard: List[Dict[str, torch.Tensor]] = [
             {} for _ in range(self.num_shards)
         ]
@@ -58,6 +73,7 @@ class PSClient:
 
     # ------------------------------------------------------------------ pulls
     def pull_dense_parameters(self, version: int = -1) -> (bool, int, Dict):
+        self._ensure_ready()
         futs = [
             c.call_future("Pserver", "pull_dense_parameters", {"version": version})
             for c in self._clients
@@ -75,6 +91,7 @@ class PSClient:
     def pull_embedding_vectors(self, name: str, ids: torch.Tensor) -> torch.Tensor:
         """Scatter unique ids to shards, gather in parallel, reorder to the
         original (possibly duplicated) id order."""
+        self._ensure_ready()
         orig_device = ids.device
         ids64 = ids.detach().to("cpu", torch.int64).reshape(-1)
         unique, inverse = torch.unique(ids64, sorted=True, return_inverse=True)
@@ -110,6 +127,7 @@ class PSClient:
     ) -> (bool, int):
         """Dense grads go to their name shard; embedding grads are merged,
         deduplicated (sum per id), then scattered by id%N."""
+        self._ensure_ready()
         per_shard_dense: List[Dict[str, torch.Tensor]] = [
             {} for _ in range(self.num_shards)
         ]

@@ -131,7 +131,7 @@ class Worker:
                 err = e
                 if attempt < 2:
                     logger.warning("minibatch failed (%s); retrying", e)
-                time.sleep(min(0.1 * attempt, 2.0))
+                time.sleep(min(0.1 * (attempt + 1), 2.0))
         raise RuntimeError(
             f"minibatch failed after {MAX_MINIBATCH_RETRY_NUM} retries"
         ) from err
@@ -219,6 +219,14 @@ class Worker:
 
 
 def _batch_len(batch) -> int:
-    if isinstance(batch, (tuple, list)):
+    """Records in a batch. Handles (x, y) tuples, plain tensors, and
+    feature-dict batches ({name: column}, y) — a dict's len() is its KEY
+    count, which silently corrupted shard accounting for feature-column
+    models (tasks never completed; the job hung in WAIT)."""
+    if isinstance(batch, dict):
+        return _batch_len(next(iter(batch.values())))
+    if isinstance(batch, (tuple, list)) and len(batch) and not isinstance(
+        batch[0], (int, float, str)
+    ):
         return _batch_len(batch[0])
     return len(batch)
