@@ -24,6 +24,7 @@ _BUILTIN = {
     "iris": "elasticdl_amd.models.iris",
     "census_wide_deep": "elasticdl_amd.models.census_wide_deep",
     "mobilenetv2": "elasticdl_amd.models.mobilenetv2",
+    "heart": "elasticdl_amd.models.heart",
 }
 
 

@@ -209,3 +209,31 @@ def test_census_end_to_end_local_job(tmp_path):
         capture_output=True, text=True, timeout=390,
     )
     assert out.returncode == 0, out.stderr[-3000:]
+
+
+def test_heart_model_trains():
+    from elasticdl_amd.models import heart as zoo
+
+    model = zoo.custom_model()
+    eng = PSEngine(opt_type="adam", opt_args="learning_rate=0.02")
+    bind_local_engine(model, eng)
+    reader = zoo.custom_data_reader("synthetic:128")
+    from elasticdl_amd.common.task import Shard, Task, TaskType
+
+    rows = list(reader.read_records(
+        Task(1, Shard(*reader.create_shards()[0]), TaskType.TRAINING)
+    ))
+    feats, labels = zoo.feed(zoo.collate_fn(rows), "cpu")
+    opt = torch.optim.Adam(model.parameters(), lr=0.02)
+    losses = []
+    for _ in range(25):
+        opt.zero_grad()
+        loss = zoo.loss(model(feats), labels)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0]
+    # CSV-shaped rows also collate
+    csv_row = ["63", "145", "233", "150", "2.3", "3", "0", "fixed", "1"]
+    f2, l2 = zoo.collate_fn([csv_row])
+    assert f2["thal"] == ["fixed"] and int(l2[0]) == 1
