@@ -195,3 +195,49 @@ def test_local_mode_subprocess_e2e():
         assert os.path.exists(export), "train-end export missing"
         state = torch.load(export, weights_only=True)
         assert any("net" in k for k in state)
+
+
+@pytest.mark.timeout(600)
+def test_ps_strategy_worker_preemption_recovers(tmp_path):
+    """PS-strategy elasticity: SIGKILL a worker mid-job; the master
+    recovers its task shards, relaunches a replacement (PS state lives
+    on), and the job completes. (Round-1 elastic evidence covered the
+    AllReduce path only.)"""
+    import signal
+    import subprocess
+    import sys
+    import time
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "mnist",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "2", "--num_ps_pods", "1",
+        "--training_data", "synthetic:4096",
+        "--minibatch_size", "32",
+        "--num_minibatches_per_task", "2",
+        "--pod_manager", "local",
+        "--device", "cpu",
+        "--checkpoint_dir", str(tmp_path),
+    ]
+    p = subprocess.Popen(
+        cmd, cwd=repo, env=dict(os.environ, PYTHONPATH=repo),
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    pidfile = os.path.join(str(tmp_path), "logs", "worker-0.pid")
+    deadline = time.time() + 120
+    wpid = None
+    while time.time() < deadline and wpid is None:
+        time.sleep(0.5)
+        if os.path.exists(pidfile):
+            with open(pidfile) as f:
+                wpid = int(f.read().strip())
+    assert wpid is not None, "worker-0 never started"
+    time.sleep(5)  # let it take tasks mid-run
+    try:
+        os.kill(wpid, signal.SIGKILL)
+    except ProcessLookupError:
+        pass  # finished early; completion still validated below
+    out, _ = p.communicate(timeout=540)
+    assert p.returncode == 0, out[-4000:]
