@@ -39,9 +39,11 @@ def resnet20_cifar(num_classes: int = 10) -> ResNet:
 
 
 def custom_model(arch: str = "cnn", **kw) -> nn.Module:
+    from elasticdl_amd.layers.batch_norm import convert_to_fused_bn
+
     if arch == "resnet":
-        return resnet20_cifar(**kw)
-    return Cifar10CNN(**kw)
+        return resnet20_cifar(**kw)  # ResNet builds fused BN itself
+    return convert_to_fused_bn(Cifar10CNN(**kw))
 
 
 def loss(outputs, labels):

@@ -96,7 +96,13 @@ class MobileNetV2(nn.Module):
 
 def custom_model(num_classes: int = 10, image_size: int = 32,
                  **kw) -> nn.Module:
-    return MobileNetV2(num_classes=num_classes, image_size=image_size, **kw)
+    from elasticdl_amd.layers.batch_norm import convert_to_fused_bn
+
+    # fused NHWC BN kernels (ReLU6 stays separate — the fused-activation
+    # path only covers plain ReLU); transparent CPU/fp32 fallback
+    return convert_to_fused_bn(
+        MobileNetV2(num_classes=num_classes, image_size=image_size, **kw)
+    )
 
 
 def loss(outputs, labels):
