@@ -50,8 +50,27 @@ __global__ __launch_bounds__(BN_THREADS) void bn_stats_kernel(
 
   float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (int64_t r = (int64_t)blockIdx.x * rpi + rl; r < R;
-       r += (int64_t)gridDim.x * rpi) {
+  const int64_t step = (int64_t)gridDim.x * rpi;
+  int64_t r = (int64_t)blockIdx.x * rpi + rl;
+  // 4x row unroll: independent in-flight loads (a single dependent
+  // load+accumulate chain leaves the memory system underfed)
+  for (; r + 3 * step < R; r += 4 * step) {
+    bn_bf16x8 v0 = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
+    bn_bf16x8 v1 =
+        *reinterpret_cast<const bn_bf16x8*>(x + (r + step) * C + c0);
+    bn_bf16x8 v2 =
+        *reinterpret_cast<const bn_bf16x8*>(x + (r + 2 * step) * C + c0);
+    bn_bf16x8 v3 =
+        *reinterpret_cast<const bn_bf16x8*>(x + (r + 3 * step) * C + c0);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float f0 = (float)v0[k], f1 = (float)v1[k];
+      float f2 = (float)v2[k], f3 = (float)v3[k];
+      s[k] += (f0 + f1) + (f2 + f3);
+      q[k] += (f0 * f0 + f1 * f1) + (f2 * f2 + f3 * f3);
+    }
+  }
+  for (; r < R; r += step) {
     bn_bf16x8 v = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
@@ -142,8 +161,28 @@ __global__ __launch_bounds__(BN_THREADS) void bn_apply_kernel(
     a[k] = rstd[c0 + k] * g;
     b[k] = (beta ? beta[c0 + k] : 0.f) - mean[c0 + k] * a[k];
   }
-  for (int64_t r = (int64_t)blockIdx.x * rpi + rl; r < R;
-       r += (int64_t)gridDim.x * rpi) {
+  const int64_t step = (int64_t)gridDim.x * rpi;
+  int64_t r = (int64_t)blockIdx.x * rpi + rl;
+  for (; r + step < R; r += 2 * step) {
+    bn_bf16x8 v0 = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
+    bn_bf16x8 v1 =
+        *reinterpret_cast<const bn_bf16x8*>(x + (r + step) * C + c0);
+    bn_bf16x8 o0, o1;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float f0 = (float)v0[k] * a[k] + b[k];
+      float f1 = (float)v1[k] * a[k] + b[k];
+      if (relu) {
+        f0 = f0 > 0.f ? f0 : 0.f;
+        f1 = f1 > 0.f ? f1 : 0.f;
+      }
+      o0[k] = (__bf16)f0;
+      o1[k] = (__bf16)f1;
+    }
+    *reinterpret_cast<bn_bf16x8*>(y + r * C + c0) = o0;
+    *reinterpret_cast<bn_bf16x8*>(y + (r + step) * C + c0) = o1;
+  }
+  for (; r < R; r += step) {
     bn_bf16x8 v = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
     bn_bf16x8 o;
 #pragma unroll
@@ -179,8 +218,28 @@ __global__ __launch_bounds__(BN_THREADS) void bn_bwd_reduce_kernel(
   }
   float s1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   float s2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (int64_t r = (int64_t)blockIdx.x * rpi + rl; r < R;
-       r += (int64_t)gridDim.x * rpi) {
+  const int64_t step = (int64_t)gridDim.x * rpi;
+  int64_t r = (int64_t)blockIdx.x * rpi + rl;
+  for (; r + step < R; r += 2 * step) {
+    bn_bf16x8 vx0 = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
+    bn_bf16x8 vd0 = *reinterpret_cast<const bn_bf16x8*>(dy + r * C + c0);
+    bn_bf16x8 vx1 =
+        *reinterpret_cast<const bn_bf16x8*>(x + (r + step) * C + c0);
+    bn_bf16x8 vd1 =
+        *reinterpret_cast<const bn_bf16x8*>(dy + (r + step) * C + c0);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float d0 = (float)vd0[k], d1 = (float)vd1[k];
+      if (relu_out != nullptr) {
+        d0 = (float)relu_out[r * C + c0 + k] > 0.f ? d0 : 0.f;
+        d1 = (float)relu_out[(r + step) * C + c0 + k] > 0.f ? d1 : 0.f;
+      }
+      s1[k] += d0 + d1;
+      s2[k] += d0 * ((float)vx0[k] - m[k]) * rs[k] +
+               d1 * ((float)vx1[k] - m[k]) * rs[k];
+    }
+  }
+  for (; r < R; r += step) {
     bn_bf16x8 vx = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
     bn_bf16x8 vd = *reinterpret_cast<const bn_bf16x8*>(dy + r * C + c0);
 #pragma unroll
@@ -285,8 +344,30 @@ __global__ __launch_bounds__(BN_THREADS) void bn_bwd_apply_kernel(
     b[k] = c2[c0 + k];
     c[k] = c3[c0 + k];
   }
-  for (int64_t r = (int64_t)blockIdx.x * rpi + rl; r < R;
-       r += (int64_t)gridDim.x * rpi) {
+  const int64_t step = (int64_t)gridDim.x * rpi;
+  int64_t r = (int64_t)blockIdx.x * rpi + rl;
+  for (; r + step < R; r += 2 * step) {
+    bn_bf16x8 vx0 = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
+    bn_bf16x8 vd0 = *reinterpret_cast<const bn_bf16x8*>(dy + r * C + c0);
+    bn_bf16x8 vx1 =
+        *reinterpret_cast<const bn_bf16x8*>(x + (r + step) * C + c0);
+    bn_bf16x8 vd1 =
+        *reinterpret_cast<const bn_bf16x8*>(dy + (r + step) * C + c0);
+    bn_bf16x8 o0, o1;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float d0 = (float)vd0[k], d1 = (float)vd1[k];
+      if (relu_out != nullptr) {
+        d0 = (float)relu_out[r * C + c0 + k] > 0.f ? d0 : 0.f;
+        d1 = (float)relu_out[(r + step) * C + c0 + k] > 0.f ? d1 : 0.f;
+      }
+      o0[k] = (__bf16)(a[k] * d0 - b[k] * (float)vx0[k] + c[k]);
+      o1[k] = (__bf16)(a[k] * d1 - b[k] * (float)vx1[k] + c[k]);
+    }
+    *reinterpret_cast<bn_bf16x8*>(dx + r * C + c0) = o0;
+    *reinterpret_cast<bn_bf16x8*>(dx + (r + step) * C + c0) = o1;
+  }
+  for (; r < R; r += step) {
     bn_bf16x8 vx = *reinterpret_cast<const bn_bf16x8*>(x + r * C + c0);
     bn_bf16x8 vd = *reinterpret_cast<const bn_bf16x8*>(dy + r * C + c0);
     bn_bf16x8 o;
@@ -363,7 +444,14 @@ void edl_add_relu_bwd(const void* dz, const void* z, void* dg, int64_t n,
 
 int edl_bn_grid_for(int64_t R, int64_t C) {
   int rpi = BN_THREADS / (int)(C >> 3);
-  return bn_grid((R + rpi - 1) / rpi);
+  int g = bn_grid((R + rpi - 1) / rpi);
+  // cap so the partial buffer (G*2C floats written + re-read by the
+  // finalize pass) stays a small fraction of the activation bytes —
+  // otherwise small-R/large-C layers pay more for partials than data
+  int64_t cap = R / 32;
+  if (cap < 64) cap = 64;
+  if (g > cap) g = (int)cap;
+  return g;
 }
 
 void edl_bn_stats(const void* x, int64_t R, int64_t C, float* part, int G,
