@@ -103,12 +103,15 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
 
         if self.track_running_stats and self.running_mean is not None:
             with torch.no_grad():
-                m = self.momentum if self.momentum is not None else 0.1
+                if self.num_batches_tracked is not None:
+                    self.num_batches_tracked += 1
+                if self.momentum is None:  # stock BN: cumulative average
+                    m = 1.0 / float(self.num_batches_tracked)
+                else:
+                    m = self.momentum
                 unbias = var * (r / max(r - 1, 1))
                 self.running_mean.mul_(1 - m).add_(mean, alpha=m)
                 self.running_var.mul_(1 - m).add_(unbias, alpha=m)
-                if self.num_batches_tracked is not None:
-                    self.num_batches_tracked += 1
 
         n, ch, h, w = x.shape
         yf = _FusedBNFn.apply(xf, self.weight, self.bias, mean, rstd,
