@@ -150,10 +150,15 @@ def add_relu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     doesn't apply (CPU, fp32, mismatched layouts)."""
     from elasticdl_amd.ops import use_native
 
+    def _dense(t):  # contiguous in SOME memory format (kernel is linear)
+        return t.is_contiguous() or (
+            t.dim() == 4 and t.is_contiguous(memory_format=torch.channels_last)
+        )
+
     if (
         a.is_cuda and a.dtype == torch.bfloat16 and b.dtype == a.dtype
         and a.stride() == b.stride() and a.numel() % 8 == 0
-        and use_native(a.device)
+        and _dense(a) and _dense(b) and use_native(a.device)
     ):
         return _AddReluFn.apply(a, b)
     return torch.relu_(a + b)
@@ -176,10 +181,9 @@ def convert_to_fused_bn(module: nn.Module) -> nn.Module:
                 track_running_stats=child.track_running_stats,
             )
             fused.load_state_dict(child.state_dict())
-            fused = fused.to(
-                device=child.weight.device if child.affine else "cpu",
-                dtype=child.weight.dtype if child.affine else None,
-            )
+            ref = child.weight if child.affine else child.running_mean
+            if ref is not None:
+                fused = fused.to(device=ref.device, dtype=ref.dtype)
             setattr(module, name, fused)
         else:
             convert_to_fused_bn(child)
