@@ -30,6 +30,16 @@ def __getattr__(name):
             "ElasticAllReduceController",
         ),
         "PSEngine": ("elasticdl_amd.ps.engine", "PSEngine"),
+        "FusedBatchNorm2d": (
+            "elasticdl_amd.layers.batch_norm", "FusedBatchNorm2d",
+        ),
+        "BNReLU": ("elasticdl_amd.layers.batch_norm", "BNReLU"),
+        "embedding_column": (
+            "elasticdl_amd.preprocessing.feature_column", "embedding_column",
+        ),
+        "DenseFeatures": (
+            "elasticdl_amd.preprocessing.feature_column", "DenseFeatures",
+        ),
     }
     if name in _exports:
         import importlib
