@@ -560,48 +560,68 @@ __global__ void ht_lookup_or_insert_kernel(
     uint8_t* __restrict__ out_is_new, int32_t* __restrict__ error_flag,
     int64_t* __restrict__ ids_by_slot) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    int64_t id = ids[i];
-    uint64_t h = edl_hash_u64((uint64_t)id) & (uint64_t)cap_mask;
+  const unsigned lane = __lane_id();
+  for (int64_t base_i =
+           (int64_t)blockIdx.x * blockDim.x + threadIdx.x - lane;
+       base_i < n; base_i += stride) {
+    int64_t i = base_i + lane;
+    int64_t id = 0;
+    int64_t claimed = -1;
     int32_t slot = -1;
-    uint8_t is_new = 0;
-    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
-      int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
-      int64_t cur = keys[pos];
-      if (cur == id) {  // inserted by a previous completed launch
-        slot = vals[pos];
-        break;
-      }
-      if (cur == EDL_EMPTY_KEY) {
-        int64_t prev = atomicCAS(reinterpret_cast<unsigned long long*>(&keys[pos]),
-                                 (unsigned long long)EDL_EMPTY_KEY,
-                                 (unsigned long long)id);
-        if (prev == EDL_EMPTY_KEY) {
-          // we claimed the key: allocate a row
-          int32_t row = atomicAdd(row_counter, 1);
-          if (row >= max_rows) {
-            atomicExch(error_flag, 1);  // arena full
-            vals[pos] = 0;
-            slot = 0;
+    if (i < n) {
+      id = ids[i];
+      uint64_t h = edl_hash_u64((uint64_t)id) & (uint64_t)cap_mask;
+      for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+        int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
+        int64_t cur = keys[pos];
+        if (cur == id) {  // inserted by a previous completed launch
+          slot = vals[pos];
+          break;
+        }
+        if (cur == EDL_EMPTY_KEY) {
+          int64_t prev = atomicCAS(
+              reinterpret_cast<unsigned long long*>(&keys[pos]),
+              (unsigned long long)EDL_EMPTY_KEY, (unsigned long long)id);
+          if (prev == EDL_EMPTY_KEY) {
+            claimed = pos;  // row allocated below (wave-aggregated)
             break;
           }
-          vals[pos] = row;
+          // lost the race to a *different* id (ids are unique per call):
+          // keep probing
+        }
+      }
+    }
+    uint8_t is_new = 0;
+    uint64_t won = __ballot(claimed >= 0);
+    if (won) {
+      int first = __ffsll((unsigned long long)won) - 1;
+      int32_t row_base = 0;
+      if ((int)lane == first)
+        row_base = atomicAdd(row_counter, __popcll((unsigned long long)won));
+      row_base = __shfl(row_base, first);
+      if (claimed >= 0) {
+        int32_t row = row_base + __popcll((unsigned long long)(
+                          won & ((1ull << lane) - 1)));
+        if (row >= max_rows) {
+          atomicExch(error_flag, 1);  // arena full
+          vals[claimed] = 0;
+          slot = 0;
+        } else {
+          vals[claimed] = row;
           if (ids_by_slot != nullptr) ids_by_slot[row] = id;
           slot = row;
           is_new = 1;
-          break;
         }
-        // lost the race to a *different* id (ids are unique per call):
-        // keep probing
       }
     }
-    if (slot < 0) {
-      atomicExch(error_flag, 2);  // table full
-      slot = 0;
+    if (i < n) {
+      if (slot < 0) {
+        atomicExch(error_flag, 2);  // table full
+        slot = 0;
+      }
+      out_slots[i] = slot;
+      if (out_is_new != nullptr) out_is_new[i] = is_new;
     }
-    out_slots[i] = slot;
-    if (out_is_new != nullptr) out_is_new[i] = is_new;
   }
 }
 
@@ -621,42 +641,67 @@ __global__ void ht_insert_dup_kernel(
     int32_t* __restrict__ error_flag,
     int64_t* __restrict__ ids_by_slot) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    int64_t id = ids[i];
-    uint64_t h = edl_hash_u64((uint64_t)id) & (uint64_t)cap_mask;
-    int32_t created = -1;
-    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
-      int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
-      int64_t cur = keys[pos];
-      if (cur == id) break;  // present (previous launch) or claimed by a peer
-      if (cur == EDL_EMPTY_KEY) {
-        int64_t prev = atomicCAS(reinterpret_cast<unsigned long long*>(&keys[pos]),
-                                 (unsigned long long)EDL_EMPTY_KEY,
-                                 (unsigned long long)id);
-        if (prev == EDL_EMPTY_KEY) {
-          int32_t row = atomicAdd(row_counter, 1);
-          if (row >= max_rows) {
-            atomicExch(error_flag, 1);
-            vals[pos] = 0;
-          } else {
-            vals[pos] = row;
-            if (ids_by_slot != nullptr) ids_by_slot[row] = id;
-            created = row;
+  const unsigned lane = __lane_id();
+  for (int64_t base_i =
+           (int64_t)blockIdx.x * blockDim.x + threadIdx.x - lane;
+       base_i < n; base_i += stride) {
+    int64_t i = base_i + lane;
+    int64_t id = 0;
+    int64_t claimed = -1;  // hash-table position this lane's CAS won
+    if (i < n) {
+      id = ids[i];
+      uint64_t h = edl_hash_u64((uint64_t)id) & (uint64_t)cap_mask;
+      for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+        int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
+        int64_t cur = keys[pos];
+        if (cur == id) break;  // present, or claimed by a peer
+        if (cur == EDL_EMPTY_KEY) {
+          int64_t prev = atomicCAS(
+              reinterpret_cast<unsigned long long*>(&keys[pos]),
+              (unsigned long long)EDL_EMPTY_KEY, (unsigned long long)id);
+          if (prev == EDL_EMPTY_KEY) {
+            claimed = pos;
+            break;
           }
-          break;
+          if (prev == id) break;  // a duplicate lane claimed it
+          // different id won this pos: keep probing
         }
-        if (prev == id) break;  // a duplicate lane claimed it
-        // different id won this pos: keep probing
       }
     }
-    if (new_slots != nullptr) new_slots[i] = created;
+    // wave-aggregated row allocation: one atomicAdd per wave of winners
+    // (zero atomics in the steady state where every id already exists)
+    int32_t created = -1;
+    uint64_t won = __ballot(claimed >= 0);
+    if (won) {
+      int first = __ffsll((unsigned long long)won) - 1;
+      int32_t row_base = 0;
+      if ((int)lane == first)
+        row_base = atomicAdd(row_counter, __popcll((unsigned long long)won));
+      row_base = __shfl(row_base, first);
+      if (claimed >= 0) {
+        int32_t row = row_base + __popcll((unsigned long long)(
+                          won & ((1ull << lane) - 1)));
+        if (row >= max_rows) {
+          atomicExch(error_flag, 1);
+          vals[claimed] = 0;
+        } else {
+          vals[claimed] = row;
+          if (ids_by_slot != nullptr) ids_by_slot[row] = id;
+          created = row;
+        }
+      }
+    }
+    if (new_slots != nullptr && i < n) new_slots[i] = created;
   }
 }
 
 // Per-batch slot compaction (slot -> dense index) with a scratch hash
 // table keyed by slot. Same duplicate-tolerant two-pass protocol:
 // pass 1 (this kernel) claims; pass 2 (batch_compact_lookup) resolves.
+// Index allocation is WAVE-AGGREGATED: per grid-stride round, the wave's
+// CAS winners take one atomicAdd for the whole wave (a single shared
+// counter address serializes ~100k per-lane atomics otherwise — measured
+// 112 us/call on DeepFM batches, the largest kernel of the PS step).
 __global__ void batch_compact_claim_kernel(
     int32_t* __restrict__ ht_keys,  // [cap] scratch, -1 = empty
     int32_t* __restrict__ ht_vals, int64_t cap_mask,
@@ -664,23 +709,42 @@ __global__ void batch_compact_claim_kernel(
     const int32_t* __restrict__ slots, int64_t n,
     int32_t* __restrict__ unique_slots /* [n] capacity */) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    int32_t slot = slots[i];
-    uint64_t h = edl_hash_u64((uint64_t)(uint32_t)slot) & (uint64_t)cap_mask;
-    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
-      int64_t pos = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
-      int32_t cur = ht_keys[pos];
-      if (cur == slot) break;
-      if (cur == -1) {
-        int32_t prev = atomicCAS(&ht_keys[pos], -1, slot);
-        if (prev == -1) {
-          int32_t idx = atomicAdd(counter, 1);
-          ht_vals[pos] = idx;
-          unique_slots[idx] = slot;
-          break;
+  const unsigned lane = __lane_id();
+  for (int64_t base_i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x - lane;
+       base_i < n; base_i += stride) {
+    int64_t i = base_i + lane;
+    int64_t pos = -1;  // claimed position (this lane won a new key)
+    int32_t slot = 0;
+    if (i < n) {
+      slot = slots[i];
+      uint64_t h =
+          edl_hash_u64((uint64_t)(uint32_t)slot) & (uint64_t)cap_mask;
+      for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+        int64_t p = (int64_t)((h + (uint64_t)probe) & (uint64_t)cap_mask);
+        int32_t cur = ht_keys[p];
+        if (cur == slot) break;
+        if (cur == -1) {
+          int32_t prev = atomicCAS(&ht_keys[p], -1, slot);
+          if (prev == -1) {
+            pos = p;
+            break;
+          }
+          if (prev == slot) break;
         }
-        if (prev == slot) break;
+      }
+    }
+    uint64_t won = __ballot(pos >= 0);
+    if (won) {
+      int first = __ffsll((unsigned long long)won) - 1;
+      int32_t idx_base = 0;
+      if ((int)lane == first)
+        idx_base = atomicAdd(counter, __popcll((unsigned long long)won));
+      idx_base = __shfl(idx_base, first);
+      if (pos >= 0) {
+        int32_t idx = idx_base + __popcll((unsigned long long)(
+                          won & ((1ull << lane) - 1)));
+        ht_vals[pos] = idx;
+        unique_slots[idx] = slot;
       }
     }
   }

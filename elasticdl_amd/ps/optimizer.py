@@ -147,8 +147,13 @@ class Optimizer:
             from elasticdl_amd.ops import _C
 
             slots_full = table.lookup_or_create_dup(ids)
-            # common case: no in-batch duplicates -> apply directly
-            if not table.has_duplicate_slots(slots_full):
+            # small batches: one device->host flag read decides whether the
+            # compaction pass is needed at all. Large CTR batches always
+            # contain duplicates — skip the detect kernel AND its .item()
+            # sync (a guaranteed pipeline bubble per push).
+            if ids.numel() < 4096 and not table.has_duplicate_slots(
+                slots_full
+            ):
                 self._apply_rows(table, grads, slots_full, lr_mult)
                 return
             unique_slots, compact_idx, u = table.compact_slots(slots_full)
