@@ -20,17 +20,18 @@ void edl_dense_adagrad(float*, float*, const float*, int64_t, float, float,
 void edl_dense_ftrl(float*, float*, float*, const float*, int64_t, float,
                     float, float, float, hipStream_t);
 void edl_sparse_sgd(float*, const float*, const int32_t*, int64_t, int64_t,
-                    float, hipStream_t);
+                    float, const int32_t*, hipStream_t);
 void edl_sparse_momentum(float*, float*, const float*, const int32_t*,
-                         int64_t, int64_t, float, float, bool, hipStream_t);
+                         int64_t, int64_t, float, float, bool,
+                         const int32_t*, hipStream_t);
 void edl_sparse_adam(float*, float*, float*, float*, const float*,
                      const int32_t*, int64_t, int64_t, float, float, float,
-                     float, hipStream_t);
+                     float, const int32_t*, hipStream_t);
 void edl_sparse_adagrad(float*, float*, const float*, const int32_t*, int64_t,
-                        int64_t, float, float, hipStream_t);
+                        int64_t, float, float, const int32_t*, hipStream_t);
 void edl_sparse_ftrl(float*, float*, float*, const float*, const int32_t*,
                      int64_t, int64_t, float, float, float, float,
-                     hipStream_t);
+                     const int32_t*, hipStream_t);
 void edl_dense_rmsprop(float*, float*, float*, float*, const float*, int64_t,
                        float, float, float, float, hipStream_t);
 void edl_dense_adadelta(float*, float*, float*, const float*, int64_t, float,
@@ -41,15 +42,16 @@ void edl_dense_nadam(float*, float*, float*, const float*, int64_t, float,
                      float, float, float, float, float, float, hipStream_t);
 void edl_sparse_rmsprop(float*, float*, float*, float*, const float*,
                         const int32_t*, int64_t, int64_t, float, float, float,
-                        float, hipStream_t);
+                        float, const int32_t*, hipStream_t);
 void edl_sparse_adadelta(float*, float*, float*, const float*, const int32_t*,
-                         int64_t, int64_t, float, float, float, hipStream_t);
+                         int64_t, int64_t, float, float, float,
+                         const int32_t*, hipStream_t);
 void edl_sparse_adamax(float*, float*, float*, const float*, const int32_t*,
                        int64_t, int64_t, float, float, float, float,
-                       hipStream_t);
+                       const int32_t*, hipStream_t);
 void edl_sparse_nadam(float*, float*, float*, const float*, const int32_t*,
                       int64_t, int64_t, float, float, float, float, float,
-                      float, float, hipStream_t);
+                      float, float, const int32_t*, hipStream_t);
 void edl_ht_lookup_or_insert(int64_t*, int32_t*, int64_t, int32_t*, int32_t,
                              const int64_t*, int64_t, int32_t*, uint8_t*,
                              int32_t*, int64_t*, hipStream_t);
@@ -203,92 +205,103 @@ static void check_sparse(const torch::Tensor& arena, const torch::Tensor& g,
 }
 
 void sparse_sgd(torch::Tensor arena, torch::Tensor g, torch::Tensor slots,
-                double lr) {
+                double lr, c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_sgd(arena.data_ptr<float>(), g.data_ptr<float>(),
                  slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr,
-                 cur_stream());
+                 live.has_value() ? live->data_ptr<int32_t>() : nullptr, cur_stream());
 }
 
 void sparse_momentum(torch::Tensor arena, torch::Tensor vel, torch::Tensor g,
-                     torch::Tensor slots, double lr, double mu, bool nesterov) {
+                     torch::Tensor slots, double lr, double mu, bool nesterov,
+                     c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_momentum(arena.data_ptr<float>(), vel.data_ptr<float>(),
                       g.data_ptr<float>(), slots.data_ptr<int32_t>(),
-                      g.size(0), g.size(1), lr, mu, nesterov, cur_stream());
+                      g.size(0), g.size(1), lr, mu, nesterov,
+                      live.has_value() ? live->data_ptr<int32_t>() : nullptr, cur_stream());
 }
 
 void sparse_adam(torch::Tensor arena, torch::Tensor m, torch::Tensor v,
                  c10::optional<torch::Tensor> max_sq, torch::Tensor g,
                  torch::Tensor slots, double lr_t, double b1, double b2,
-                 double eps) {
+                 double eps, c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_adam(arena.data_ptr<float>(), m.data_ptr<float>(),
                   v.data_ptr<float>(),
                   max_sq.has_value() ? max_sq->data_ptr<float>() : nullptr,
                   g.data_ptr<float>(), slots.data_ptr<int32_t>(), g.size(0),
-                  g.size(1), lr_t, b1, b2, eps, cur_stream());
+                  g.size(1), lr_t, b1, b2, eps, live.has_value() ? live->data_ptr<int32_t>() : nullptr,
+                  cur_stream());
 }
 
 void sparse_adagrad(torch::Tensor arena, torch::Tensor m, torch::Tensor g,
-                    torch::Tensor slots, double lr, double eps) {
+                    torch::Tensor slots, double lr, double eps,
+                    c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_adagrad(arena.data_ptr<float>(), m.data_ptr<float>(),
                      g.data_ptr<float>(), slots.data_ptr<int32_t>(), g.size(0),
-                     g.size(1), lr, eps, cur_stream());
+                     g.size(1), lr, eps, live.has_value() ? live->data_ptr<int32_t>() : nullptr,
+                     cur_stream());
 }
 
 void sparse_ftrl(torch::Tensor arena, torch::Tensor z, torch::Tensor n,
                  torch::Tensor g, torch::Tensor slots, double alpha,
-                 double beta, double l1, double l2) {
+                 double beta, double l1, double l2,
+                 c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_ftrl(arena.data_ptr<float>(), z.data_ptr<float>(),
                   n.data_ptr<float>(), g.data_ptr<float>(),
                   slots.data_ptr<int32_t>(), g.size(0), g.size(1), alpha, beta,
-                  l1, l2, cur_stream());
+                  l1, l2, live.has_value() ? live->data_ptr<int32_t>() : nullptr, cur_stream());
 }
 
 void sparse_rmsprop(torch::Tensor arena, torch::Tensor ms, torch::Tensor mom,
                     c10::optional<torch::Tensor> mg, torch::Tensor g,
                     torch::Tensor slots, double lr, double rho,
-                    double momentum, double eps) {
+                    double momentum, double eps,
+                    c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_rmsprop(arena.data_ptr<float>(), ms.data_ptr<float>(),
                      mom.data_ptr<float>(),
                      mg.has_value() ? mg->data_ptr<float>() : nullptr,
                      g.data_ptr<float>(), slots.data_ptr<int32_t>(),
                      g.size(0), g.size(1), lr, rho, momentum, eps,
-                     cur_stream());
+                     live.has_value() ? live->data_ptr<int32_t>() : nullptr, cur_stream());
 }
 
 void sparse_adadelta(torch::Tensor arena, torch::Tensor ag, torch::Tensor au,
                      torch::Tensor g, torch::Tensor slots, double lr,
-                     double rho, double eps) {
+                     double rho, double eps,
+                     c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_adadelta(arena.data_ptr<float>(), ag.data_ptr<float>(),
                       au.data_ptr<float>(), g.data_ptr<float>(),
                       slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr,
-                      rho, eps, cur_stream());
+                      rho, eps, live.has_value() ? live->data_ptr<int32_t>() : nullptr, cur_stream());
 }
 
 void sparse_adamax(torch::Tensor arena, torch::Tensor m, torch::Tensor v,
                    torch::Tensor g, torch::Tensor slots, double lr_t,
-                   double b1, double b2, double eps) {
+                   double b1, double b2, double eps,
+                   c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_adamax(arena.data_ptr<float>(), m.data_ptr<float>(),
                     v.data_ptr<float>(), g.data_ptr<float>(),
                     slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr_t, b1,
-                    b2, eps, cur_stream());
+                    b2, eps, live.has_value() ? live->data_ptr<int32_t>() : nullptr, cur_stream());
 }
 
 void sparse_nadam(torch::Tensor arena, torch::Tensor m, torch::Tensor v,
                   torch::Tensor g, torch::Tensor slots, double lr, double c1,
-                  double c2, double vcorr, double b1, double b2, double eps) {
+                  double c2, double vcorr, double b1, double b2, double eps,
+                  c10::optional<torch::Tensor> live) {
   check_sparse(arena, g, slots);
   edl_sparse_nadam(arena.data_ptr<float>(), m.data_ptr<float>(),
                    v.data_ptr<float>(), g.data_ptr<float>(),
                    slots.data_ptr<int32_t>(), g.size(0), g.size(1), lr, c1,
-                   c2, vcorr, b1, b2, eps, cur_stream());
+                   c2, vcorr, b1, b2, eps, live.has_value() ? live->data_ptr<int32_t>() : nullptr,
+                   cur_stream());
 }
 
 // ------------------------------ hash table ------------------------------
@@ -564,19 +577,42 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dense_adam", &dense_adam);
   m.def("dense_adagrad", &dense_adagrad);
   m.def("dense_ftrl", &dense_ftrl);
-  m.def("sparse_sgd", &sparse_sgd);
-  m.def("sparse_momentum", &sparse_momentum);
-  m.def("sparse_adam", &sparse_adam);
-  m.def("sparse_adagrad", &sparse_adagrad);
-  m.def("sparse_ftrl", &sparse_ftrl);
+  m.def("sparse_sgd", &sparse_sgd, py::arg("arena"), py::arg("g"),
+        py::arg("slots"), py::arg("lr"), py::arg("live") = py::none());
+  m.def("sparse_momentum", &sparse_momentum, py::arg("arena"),
+        py::arg("vel"), py::arg("g"), py::arg("slots"), py::arg("lr"),
+        py::arg("mu"), py::arg("nesterov"), py::arg("live") = py::none());
+  m.def("sparse_adam", &sparse_adam, py::arg("arena"), py::arg("m"),
+        py::arg("v"), py::arg("max_sq"), py::arg("g"), py::arg("slots"),
+        py::arg("lr_t"), py::arg("b1"), py::arg("b2"), py::arg("eps"),
+        py::arg("live") = py::none());
+  m.def("sparse_adagrad", &sparse_adagrad, py::arg("arena"), py::arg("m"),
+        py::arg("g"), py::arg("slots"), py::arg("lr"), py::arg("eps"),
+        py::arg("live") = py::none());
+  m.def("sparse_ftrl", &sparse_ftrl, py::arg("arena"), py::arg("z"),
+        py::arg("n"), py::arg("g"), py::arg("slots"), py::arg("alpha"),
+        py::arg("beta"), py::arg("l1"), py::arg("l2"),
+        py::arg("live") = py::none());
   m.def("dense_rmsprop", &dense_rmsprop);
   m.def("dense_adadelta", &dense_adadelta);
   m.def("dense_adamax", &dense_adamax);
   m.def("dense_nadam", &dense_nadam);
-  m.def("sparse_rmsprop", &sparse_rmsprop);
-  m.def("sparse_adadelta", &sparse_adadelta);
-  m.def("sparse_adamax", &sparse_adamax);
-  m.def("sparse_nadam", &sparse_nadam);
+  m.def("sparse_rmsprop", &sparse_rmsprop, py::arg("arena"), py::arg("ms"),
+        py::arg("mom"), py::arg("mg"), py::arg("g"), py::arg("slots"),
+        py::arg("lr"), py::arg("rho"), py::arg("momentum"), py::arg("eps"),
+        py::arg("live") = py::none());
+  m.def("sparse_adadelta", &sparse_adadelta, py::arg("arena"),
+        py::arg("ag"), py::arg("au"), py::arg("g"), py::arg("slots"),
+        py::arg("lr"), py::arg("rho"), py::arg("eps"),
+        py::arg("live") = py::none());
+  m.def("sparse_adamax", &sparse_adamax, py::arg("arena"), py::arg("m"),
+        py::arg("v"), py::arg("g"), py::arg("slots"), py::arg("lr_t"),
+        py::arg("b1"), py::arg("b2"), py::arg("eps"),
+        py::arg("live") = py::none());
+  m.def("sparse_nadam", &sparse_nadam, py::arg("arena"), py::arg("m"),
+        py::arg("v"), py::arg("g"), py::arg("slots"), py::arg("lr"),
+        py::arg("c1"), py::arg("c2"), py::arg("vcorr"), py::arg("b1"),
+        py::arg("b2"), py::arg("eps"), py::arg("live") = py::none());
   m.def("ht_lookup_or_insert", &ht_lookup_or_insert);
   m.def("ht_lookup", &ht_lookup);
   m.def("ht_insert_dup", &ht_insert_dup);

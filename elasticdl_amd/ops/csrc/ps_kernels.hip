@@ -355,12 +355,18 @@ __global__ void dense_update_kernel(Op op, const float* __restrict__ grads,
   }
 }
 
+// ``live``: optional DEVICE row count (<= n). The sync-free sparse push
+// compacts duplicates into a device counter and launches the update over
+// the worst-case n rows; lanes beyond *live exit. This keeps the whole
+// push pipeline free of device->host reads (no .item() bubbles).
 template <typename Op>
 __global__ void sparse_update_vec_kernel(Op op, const float* __restrict__ grads,
                                          const int32_t* __restrict__ slots,
-                                         int64_t n, int64_t dim) {
+                                         int64_t n, int64_t dim,
+                                         const int32_t* __restrict__ live) {
   int64_t dim4 = dim >> 2;
-  int64_t total = n * dim4;
+  int64_t nn = live != nullptr ? (int64_t)*live : n;
+  int64_t total = nn * dim4;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
@@ -376,8 +382,10 @@ template <typename Op>
 __global__ void sparse_update_scalar_kernel(Op op,
                                             const float* __restrict__ grads,
                                             const int32_t* __restrict__ slots,
-                                            int64_t n, int64_t dim) {
-  int64_t total = n * dim;
+                                            int64_t n, int64_t dim,
+                                            const int32_t* __restrict__ live) {
+  int64_t nn = live != nullptr ? (int64_t)*live : n;
+  int64_t total = nn * dim;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
@@ -397,14 +405,16 @@ static void launch_dense(Op op, const float* grads, int64_t numel,
 
 template <typename Op>
 static void launch_sparse(Op op, const float* grads, const int32_t* slots,
-                          int64_t n, int64_t dim, hipStream_t stream) {
+                          int64_t n, int64_t dim, hipStream_t stream,
+                          const int32_t* live = nullptr) {
   if ((dim & 3) == 0) {
     sparse_update_vec_kernel<Op>
         <<<grid_for(n * (dim >> 2)), THREADS, 0, stream>>>(op, grads, slots, n,
-                                                           dim);
+                                                           dim, live);
   } else {
     sparse_update_scalar_kernel<Op>
-        <<<grid_for(n * dim), THREADS, 0, stream>>>(op, grads, slots, n, dim);
+        <<<grid_for(n * dim), THREADS, 0, stream>>>(op, grads, slots, n, dim,
+                                                    live);
   }
 }
 
@@ -465,64 +475,71 @@ void edl_dense_nadam(float* p, float* m, float* v, const float* g,
 }
 
 void edl_sparse_sgd(float* arena, const float* g, const int32_t* slots,
-                    int64_t n, int64_t dim, float lr, hipStream_t s) {
-  launch_sparse(SgdOp{arena, lr}, g, slots, n, dim, s);
+                    int64_t n, int64_t dim, float lr,
+                    const int32_t* live, hipStream_t s) {
+  launch_sparse(SgdOp{arena, lr}, g, slots, n, dim, s, live);
 }
 
 void edl_sparse_momentum(float* arena, float* vel, const float* g,
                          const int32_t* slots, int64_t n, int64_t dim,
-                         float lr, float mu, bool nesterov, hipStream_t s) {
-  launch_sparse(MomentumOp{arena, vel, lr, mu, nesterov}, g, slots, n, dim, s);
+                         float lr, float mu, bool nesterov,
+                         const int32_t* live, hipStream_t s) {
+  launch_sparse(MomentumOp{arena, vel, lr, mu, nesterov}, g, slots, n, dim, s,
+                live);
 }
 
 void edl_sparse_adam(float* arena, float* m, float* v, float* max_sq,
                      const float* g, const int32_t* slots, int64_t n,
                      int64_t dim, float lr_t, float b1, float b2, float eps,
-                     hipStream_t s) {
+                     const int32_t* live, hipStream_t s) {
   launch_sparse(AdamOp{arena, m, v, max_sq, lr_t, b1, b2, eps}, g, slots, n,
-                dim, s);
+                dim, s, live);
 }
 
 void edl_sparse_adagrad(float* arena, float* m, const float* g,
                         const int32_t* slots, int64_t n, int64_t dim, float lr,
-                        float eps, hipStream_t s) {
-  launch_sparse(AdagradOp{arena, m, lr, eps}, g, slots, n, dim, s);
+                        float eps, const int32_t* live, hipStream_t s) {
+  launch_sparse(AdagradOp{arena, m, lr, eps}, g, slots, n, dim, s, live);
 }
 
 void edl_sparse_ftrl(float* arena, float* z, float* nacc, const float* g,
                      const int32_t* slots, int64_t n, int64_t dim, float alpha,
-                     float beta, float l1, float l2, hipStream_t s) {
+                     float beta, float l1, float l2, const int32_t* live,
+                     hipStream_t s) {
   launch_sparse(FtrlOp{arena, z, nacc, alpha, beta, l1, l2}, g, slots, n, dim,
-                s);
+                s, live);
 }
 
 void edl_sparse_rmsprop(float* arena, float* ms, float* mom, float* mg,
                         const float* g, const int32_t* slots, int64_t n,
                         int64_t dim, float lr, float rho, float momentum,
-                        float eps, hipStream_t s) {
+                        float eps, const int32_t* live, hipStream_t s) {
   launch_sparse(RmspropOp{arena, ms, mom, mg, lr, rho, momentum, eps}, g,
-                slots, n, dim, s);
+                slots, n, dim, s, live);
 }
 
 void edl_sparse_adadelta(float* arena, float* ag, float* au, const float* g,
                          const int32_t* slots, int64_t n, int64_t dim,
-                         float lr, float rho, float eps, hipStream_t s) {
-  launch_sparse(AdadeltaOp{arena, ag, au, lr, rho, eps}, g, slots, n, dim, s);
+                         float lr, float rho, float eps,
+                         const int32_t* live, hipStream_t s) {
+  launch_sparse(AdadeltaOp{arena, ag, au, lr, rho, eps}, g, slots, n, dim, s,
+                live);
 }
 
 void edl_sparse_adamax(float* arena, float* m, float* v, const float* g,
                        const int32_t* slots, int64_t n, int64_t dim,
                        float lr_t, float b1, float b2, float eps,
-                       hipStream_t s) {
-  launch_sparse(AdamaxOp{arena, m, v, lr_t, b1, b2, eps}, g, slots, n, dim, s);
+                       const int32_t* live, hipStream_t s) {
+  launch_sparse(AdamaxOp{arena, m, v, lr_t, b1, b2, eps}, g, slots, n, dim, s,
+                live);
 }
 
 void edl_sparse_nadam(float* arena, float* m, float* v, const float* g,
                       const int32_t* slots, int64_t n, int64_t dim, float lr,
                       float c1, float c2, float vcorr, float b1, float b2,
-                      float eps, hipStream_t s) {
+                      float eps, const int32_t* live, hipStream_t s) {
   launch_sparse(NadamOp{arena, m, v, lr, c1, c2, vcorr, b1, b2, eps}, g, slots,
-                n, dim, s);
+                n, dim, s, live);
 }
 
 }  // extern "C"

@@ -156,18 +156,24 @@ class Optimizer:
             ):
                 self._apply_rows(table, grads, slots_full, lr_mult)
                 return
-            unique_slots, compact_idx, u = table.compact_slots(slots_full)
+            # fully sync-free compacted path: the unique count stays on
+            # device; acc is sized for the worst case and the counted
+            # sparse kernels skip rows beyond *u_dev
+            unique_slots, compact_idx, u_dev = table.compact_slots_async(
+                slots_full
+            )
             acc = torch.zeros(
-                (u, table.dim), dtype=torch.float32, device=table.device
+                (ids.numel(), table.dim), dtype=torch.float32,
+                device=table.device,
             )
             _C.accumulate_rows(grads, compact_idx, acc)
-            self._apply_rows(table, acc, unique_slots.contiguous(), lr_mult)
+            self._apply_rows(table, acc, unique_slots, lr_mult, live=u_dev)
             return
         summed, unique_ids = deduplicate_indexed_slices(grads, ids)
         slots = table.lookup_or_create(unique_ids)
         self._apply_rows(table, summed.contiguous(), slots, lr_mult)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         raise NotImplementedError
 
 
@@ -181,12 +187,12 @@ class SGDOptimizer(Optimizer):
         else:
             reference.dense_sgd(param, grad, lr)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         lr = self.base_lr * lr_mult
         if self._native(table.arena):
             from elasticdl_amd.ops import _C
 
-            _C.sparse_sgd(table.arena, grads, slots, lr)
+            _C.sparse_sgd(table.arena, grads, slots, lr, live)
         else:
             reference.sparse_sgd(table.arena, grads, slots, lr)
 
@@ -209,13 +215,14 @@ class MomentumOptimizer(Optimizer):
         else:
             reference.dense_momentum(param, vel, grad, lr, self.mu, self.nesterov)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         vel = table.get_slot_arena("momentum")
         lr = self.base_lr * lr_mult
         if self._native(table.arena):
             from elasticdl_amd.ops import _C
 
-            _C.sparse_momentum(table.arena, vel, grads, slots, lr, self.mu, self.nesterov)
+            _C.sparse_momentum(table.arena, vel, grads, slots, lr, self.mu,
+                               self.nesterov, live)
         else:
             reference.sparse_momentum(table.arena, vel, grads, slots, lr, self.mu, self.nesterov)
 
@@ -247,7 +254,7 @@ class AdamOptimizer(Optimizer):
             reference.dense_adam(param, st["m"], st["v"], ms, grad,
                                  lr_t, self.b1, self.b2, self.eps)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         m = table.get_slot_arena("m")
         v = table.get_slot_arena("v")
         ms = table.get_slot_arena("max_square") if self.amsgrad else None
@@ -256,7 +263,7 @@ class AdamOptimizer(Optimizer):
             from elasticdl_amd.ops import _C
 
             _C.sparse_adam(table.arena, m, v, ms, grads, slots,
-                           lr_t, self.b1, self.b2, self.eps)
+                           lr_t, self.b1, self.b2, self.eps, live)
         else:
             reference.sparse_adam(table.arena, m, v, ms, grads, slots,
                                   lr_t, self.b1, self.b2, self.eps)
@@ -279,13 +286,13 @@ class AdagradOptimizer(Optimizer):
         else:
             reference.dense_adagrad(param, m, grad, lr, self.eps)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         m = table.get_slot_arena("accumulator")
         lr = self.base_lr * lr_mult
         if self._native(table.arena):
             from elasticdl_amd.ops import _C
 
-            _C.sparse_adagrad(table.arena, m, grads, slots, lr, self.eps)
+            _C.sparse_adagrad(table.arena, m, grads, slots, lr, self.eps, live)
         else:
             reference.sparse_adagrad(table.arena, m, grads, slots, lr, self.eps)
 
@@ -319,7 +326,7 @@ class RmspropOptimizer(Optimizer):
                                     grad, lr, self.rho, self.momentum,
                                     self.eps)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         ms = table.get_slot_arena("rms")
         mom = table.get_slot_arena("momentum")
         mg = table.get_slot_arena("mg") if self.centered else None
@@ -328,7 +335,7 @@ class RmspropOptimizer(Optimizer):
             from elasticdl_amd.ops import _C
 
             _C.sparse_rmsprop(table.arena, ms, mom, mg, grads, slots, lr,
-                              self.rho, self.momentum, self.eps)
+                              self.rho, self.momentum, self.eps, live)
         else:
             reference.sparse_rmsprop(table.arena, ms, mom, mg, grads, slots,
                                      lr, self.rho, self.momentum, self.eps)
@@ -354,7 +361,7 @@ class AdadeltaOptimizer(Optimizer):
                                      st["accum_var"], grad, lr, self.rho,
                                      self.eps)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         ag = table.get_slot_arena("accum_grad")
         au = table.get_slot_arena("accum_var")
         lr = self.base_lr * lr_mult
@@ -362,7 +369,7 @@ class AdadeltaOptimizer(Optimizer):
             from elasticdl_amd.ops import _C
 
             _C.sparse_adadelta(table.arena, ag, au, grads, slots, lr,
-                               self.rho, self.eps)
+                               self.rho, self.eps, live)
         else:
             reference.sparse_adadelta(table.arena, ag, au, grads, slots, lr,
                                       self.rho, self.eps)
@@ -392,7 +399,7 @@ class AdamaxOptimizer(Optimizer):
             reference.dense_adamax(param, st["m"], st["v"], grad, lr_t,
                                    self.b1, self.b2, self.eps)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         m = table.get_slot_arena("m")
         v = table.get_slot_arena("v")
         lr_t = self._lr_t(lr_mult)
@@ -400,7 +407,7 @@ class AdamaxOptimizer(Optimizer):
             from elasticdl_amd.ops import _C
 
             _C.sparse_adamax(table.arena, m, v, grads, slots, lr_t, self.b1,
-                             self.b2, self.eps)
+                             self.b2, self.eps, live)
         else:
             reference.sparse_adamax(table.arena, m, v, grads, slots, lr_t,
                                     self.b1, self.b2, self.eps)
@@ -431,7 +438,7 @@ class NadamOptimizer(Optimizer):
             reference.dense_nadam(param, st["m"], st["v"], grad, lr, c1, c2,
                                   vcorr, self.b1, self.b2, self.eps)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         m = table.get_slot_arena("m")
         v = table.get_slot_arena("v")
         c1, c2, vcorr = self._coeffs()
@@ -440,7 +447,7 @@ class NadamOptimizer(Optimizer):
             from elasticdl_amd.ops import _C
 
             _C.sparse_nadam(table.arena, m, v, grads, slots, lr, c1, c2,
-                            vcorr, self.b1, self.b2, self.eps)
+                            vcorr, self.b1, self.b2, self.eps, live)
         else:
             reference.sparse_nadam(table.arena, m, v, grads, slots, lr, c1,
                                    c2, vcorr, self.b1, self.b2, self.eps)
@@ -465,7 +472,7 @@ class FtrlOptimizer(Optimizer):
             reference.dense_ftrl(param, st["linear"], st["accumulator"],
                                  grad, alpha, self.beta, self.l1, self.l2)
 
-    def _apply_rows(self, table, grads, slots, lr_mult):
+    def _apply_rows(self, table, grads, slots, lr_mult, live=None):
         z = table.get_slot_arena("linear")
         n = table.get_slot_arena("accumulator")
         alpha = self.base_lr * lr_mult
@@ -473,7 +480,7 @@ class FtrlOptimizer(Optimizer):
             from elasticdl_amd.ops import _C
 
             _C.sparse_ftrl(table.arena, z, n, grads, slots,
-                           alpha, self.beta, self.l1, self.l2)
+                           alpha, self.beta, self.l1, self.l2, live)
         else:
             reference.sparse_ftrl(table.arena, z, n, grads, slots,
                                   alpha, self.beta, self.l1, self.l2)

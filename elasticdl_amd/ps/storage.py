@@ -228,9 +228,11 @@ class EmbeddingTable:
         self._C.detect_dup_slots(slots, self._mark, self._mark_tag, self._dup_flag)
         return bool(self._dup_flag.item())
 
-    def compact_slots(self, slots: torch.Tensor):
-        """(unique_slots [u], compact_idx [n], u) via the batch scratch
-        hash — GPU only."""
+    def compact_slots_async(self, slots: torch.Tensor):
+        """(unique_slots [n, valid prefix], compact_idx [n], u_dev) via the
+        batch scratch hash — GPU only, NO device->host sync: ``u_dev`` is
+        the device-resident unique count consumed by the counted sparse
+        kernels (rows beyond *u_dev are skipped in-kernel)."""
         assert self._native
         n = slots.numel()
         cap = _next_pow2(max(2 * n, 16))
@@ -248,7 +250,12 @@ class EmbeddingTable:
             self._bc_keys, self._bc_vals, self._bc_counter, slots,
             unique_slots, compact_idx,
         )
-        u = int(self._bc_counter.item())
+        return unique_slots, compact_idx, self._bc_counter
+
+    def compact_slots(self, slots: torch.Tensor):
+        """(unique_slots [u], compact_idx [n], u) — synchronous variant."""
+        unique_slots, compact_idx, u_dev = self.compact_slots_async(slots)
+        u = int(u_dev.item())
         return unique_slots[:u], compact_idx, u
 
     def lookup(self, ids: torch.Tensor) -> torch.Tensor:

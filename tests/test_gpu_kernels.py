@@ -500,3 +500,36 @@ def test_init_normal_statistics(C):
     vals = arena.cpu().reshape(-1)
     assert vals.abs().max().item() <= 2.0 + 1e-6
     assert vals.std().item() < 0.95
+
+
+def test_counted_sparse_path_matches_sync_path(C):
+    """Sync-free compacted apply (device-resident unique count) must
+    produce the same arena as the synchronous dedup path."""
+    from elasticdl_amd.common.tensor_utils import IndexedSlices
+    from elasticdl_amd.ps.engine import PSEngine
+
+    ids = torch.cat([
+        torch.randint(0, 5000, (8192,)),   # large, duplicate-heavy
+        torch.tensor([1, 1, 1, 2]),
+    ])
+    grads = torch.randn(ids.numel(), 16, device="cuda")
+
+    eng = PSEngine(opt_type="adam", opt_args="learning_rate=0.01",
+                   device="cuda")
+    eng.push_model({}, [{"name": "t", "dim": 16}])
+    eng.push_gradients({}, {"t": IndexedSlices(grads, ids.cuda())})
+
+    eng2 = PSEngine(opt_type="adam", opt_args="learning_rate=0.01",
+                    device="cuda")
+    eng2.push_model({}, [{"name": "t", "dim": 16}])
+    # oracle: CPU-style unique+sum then unique-slot apply on GPU
+    from elasticdl_amd.common.tensor_utils import deduplicate_indexed_slices
+
+    summed, uids = deduplicate_indexed_slices(grads, ids.cuda())
+    eng2.push_gradients({}, {"t": IndexedSlices(summed, uids)})
+
+    torch.cuda.synchronize()
+    probe = torch.unique(ids)[:512]
+    r1 = eng.pull_embedding_vectors("t", probe)
+    r2 = eng2.pull_embedding_vectors("t", probe)
+    assert torch.allclose(r1, r2, atol=1e-5), (r1 - r2).abs().max()
