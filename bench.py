@@ -248,6 +248,14 @@ def bench_ps_model(args, rank, world, local_rank):
         and world == 1
     )
     if use_graphs:
+        # documented capture protocol: warm up on a side stream first so
+        # autograd/allocator state is steady before recording
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for i in range(2 * len(batches)):
+                one_step(i)
+        torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         graphs = []
         for i in range(len(batches)):
