@@ -235,40 +235,6 @@ def bench_ps_model(args, rank, world, local_rank):
     for i in range(args.warmup):
         one_step(i)
 
-    # Opt-in hipGraph capture (EDL_HIP_GRAPHS=1, single GPU): the sparse
-    # push is fully sync-free, so the whole fwd+bwd+dense-step+push
-    # records as one graph per resident batch and replays with a single
-    # launch. NOT used for the official numbers (host-side bookkeeping —
-    # version counters, Adam bias-correction scalars — is frozen at its
-    # captured value on replay); reported separately in
-    # profiles/ps_path_r02.md as the production launch-bound ceiling.
-    use_graphs = (
-        os.environ.get("EDL_HIP_GRAPHS") == "1"
-        and device.type == "cuda"
-        and world == 1
-    )
-    if use_graphs:
-        # documented capture protocol: warm up on a side stream first so
-        # autograd/allocator state is steady before recording
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            for i in range(2 * len(batches)):
-                one_step(i)
-        torch.cuda.current_stream().wait_stream(side)
-        torch.cuda.synchronize()
-        graphs = []
-        for i in range(len(batches)):
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                one_step(i)
-            graphs.append(g)
-
-        def one_step(i, _graphs=graphs):  # noqa: F811 - replay variant
-            _graphs[i % len(_graphs)].replay()
-
-        one_step(0)  # replay warmup
-
     sync_all(dist, world, device)
     t0 = time.perf_counter()
     for i in range(args.steps):
