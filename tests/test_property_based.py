@@ -74,3 +74,44 @@ def test_task_shard_math_covers_all_records(shards, rpt):
         total += t.shard.size
         tm.report(t.task_id, True, 0)
     assert total == sum(b - a for _, a, b in shards)
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    st.lists(st.binary(min_size=0, max_size=200), min_size=0, max_size=60),
+    st.integers(min_value=32, max_value=512),
+    st.sampled_from([0, 2]),  # COMPRESS_NONE / COMPRESS_DEFLATE
+)
+def test_recordio_round_trip_property(records, chunk_bytes, compressor):
+    import tempfile
+
+    from elasticdl_amd.data.recordio import Index, Scanner, Writer
+
+    with tempfile.TemporaryDirectory() as d:
+        path = f"{d}/f.recordio"
+        with Writer(path, max_chunk_bytes=chunk_bytes,
+                    compressor=compressor) as w:
+            for r in records:
+                w.write(r)
+        assert Index(path).num_records() == len(records)
+        assert list(Scanner(path)) == records
+        # arbitrary subranges
+        if records:
+            start = len(records) // 2
+            assert list(Scanner(path, start, 3)) == records[start:start + 3]
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.lists(st.lists(st.integers(0, 1000), max_size=6), min_size=1,
+                max_size=8))
+def test_to_ragged_to_sparse_round_trip(rows):
+    from elasticdl_amd.preprocessing.layers import PAD, ToRagged, ToSparse
+
+    padded = ToRagged()(rows)
+    assert padded.shape[0] == len(rows)
+    for i, row in enumerate(rows):
+        assert padded[i, :len(row)].tolist() == row
+        assert (padded[i, len(row):] == PAD).all()
+    sp = ToSparse()(padded)
+    nnz_expected = sum(1 for r in rows for v in r if v != PAD)
+    assert sp._nnz() == nnz_expected
