@@ -166,8 +166,13 @@ class ParameterServerTrainer(Trainer):
             raise RuntimeError("gradients rejected as stale; re-pulled model")
         if self._local_opt is not None and self._local_step % self.get_model_steps:
             self._local_opt.step()
-        self._version = max(self._version, version)
-        return loss.detach(), self._version
+        # NOTE: self._version is the version of the LOCAL weights and only
+        # advances on an actual pull. Claiming the post-push version here
+        # would satisfy the PS's version gate forever, so the worker would
+        # never receive its own updates back — a 1-worker async job then
+        # trains every step against the initial weights and learns nothing
+        # (found by the PS-mode convergence bench; regression-tested).
+        return loss.detach(), version
 
     def _feed(self, batch):
         if self.spec.feed_fn is not None:

@@ -45,6 +45,13 @@ def loss(outputs, labels):
 
 
 def optimizer(model=None):
+    import os
+
+    # async PS updates interleave between pulls: momentum compounds
+    # across stale applications and diverges (true for the reference's
+    # async mode too) — use the async-appropriate plain-SGD setting
+    if os.environ.get("EDL_CONV_STRATEGY") == "ps":
+        return ("sgd", "learning_rate=0.03;momentum=0.0")
     return ("sgd", "learning_rate=0.05;momentum=0.9")
 
 
@@ -73,14 +80,16 @@ def custom_data_reader(data_origin=""):
 '''
 
 
-def run(num_workers, kill_one=False, records=8192):
+def run(num_workers, kill_one=False, records=8192, strategy="allreduce"):
     with open(ZOO, "w") as f:
         f.write(ZOO_SRC)
     tmp = tempfile.mkdtemp(prefix="edl-conv-")
+    strat = ("ParameterServerStrategy" if strategy == "ps"
+             else "AllreduceStrategy")
     cmd = [
         sys.executable, "-m", "elasticdl_amd.master.main",
         "--model_def", ZOO,
-        "--distribution_strategy", "AllreduceStrategy",
+        "--distribution_strategy", strat,
         "--num_workers", str(num_workers),
         "--minibatch_size", "32",
         "--num_minibatches_per_task", "4",
@@ -91,8 +100,15 @@ def run(num_workers, kill_one=False, records=8192):
         "--log_loss_steps", "5",
         "--pod_manager", "local",
     ]
+    if strategy == "ps":
+        cmd += ["--num_ps_pods", "1", "--use_async", "true"]
+        env_extra = {"EDL_CONV_STRATEGY": "ps"}
+    else:
+        env_extra = {}
     env = dict(os.environ, PYTHONPATH=REPO, EDL_PG_TIMEOUT_SEC="20",
-               EDL_MIN_WORLD=str(num_workers))
+               **env_extra)
+    if strategy != "ps":
+        env["EDL_MIN_WORLD"] = str(num_workers)
     p = subprocess.Popen(cmd, env=env, cwd=REPO, stdout=subprocess.PIPE,
                          stderr=subprocess.STDOUT, text=True)
     if kill_one:
@@ -126,19 +142,27 @@ def run(num_workers, kill_one=False, records=8192):
 
 
 def main():
-    results = {}
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--strategy", default="allreduce",
+                    choices=["allreduce", "ps"])
+    args = ap.parse_args()
+    results = {"strategy": args.strategy}
     for name, kw in [
-        ("fixed_2_workers", dict(num_workers=2)),
-        ("fixed_1_worker", dict(num_workers=1)),
-        ("elastic_2_minus_1", dict(num_workers=2, kill_one=True)),
+        ("fixed_2_workers", dict(num_workers=2, strategy=args.strategy)),
+        ("fixed_1_worker", dict(num_workers=1, strategy=args.strategy)),
+        ("elastic_2_minus_1",
+         dict(num_workers=2, kill_one=True, strategy=args.strategy)),
     ]:
         rc, final, n = run(**kw)
         results[name] = {"exit": rc, "final_loss": final, "logged_steps": n}
         print(f"[convergence] {name}: {results[name]}", flush=True)
     print(json.dumps({"metric": "convergence_parity", **results}), flush=True)
     ok = all(v["exit"] == 0 and v["final_loss"] is not None
-             for v in results.values())
-    losses = [v["final_loss"] for v in results.values()]
+             for k, v in results.items() if k != "strategy")
+    losses = [v["final_loss"] for k, v in results.items()
+              if k != "strategy" and v["final_loss"] is not None]
     spread = max(losses) - min(losses)
     print(f"[convergence] final-loss spread across runs: {spread:.4f}",
           flush=True)

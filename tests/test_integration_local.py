@@ -241,3 +241,65 @@ def test_ps_strategy_worker_preemption_recovers(tmp_path):
         pass  # finished early; completion still validated below
     out, _ = p.communicate(timeout=540)
     assert p.returncode == 0, out[-4000:]
+
+
+def test_single_worker_ps_actually_learns():
+    """Regression: after a push the trainer must NOT claim the post-push
+    model version — that satisfied the PS's pull version-gate forever,
+    so a 1-worker async job computed every gradient against the initial
+    weights and never learned (loss pinned at ln 2 on separable data)."""
+    import types
+
+    import torch.nn as nn
+
+    from elasticdl_amd.data.reader import SyntheticReader
+    from elasticdl_amd.utils.model_utils import ModelSpec
+
+    torch.manual_seed(0)
+    w_true = torch.randn(16)
+
+    def sample(i):
+        g = torch.Generator().manual_seed(i)
+        x = torch.randn(16, generator=g)
+        return x, (x @ w_true > 0).long()
+
+    reader = SyntheticReader(2048, sample, records_per_shard=128)
+
+    def model_fn():
+        torch.manual_seed(7)
+        return nn.Sequential(nn.Linear(16, 64), nn.ReLU(),
+                             nn.Linear(64, 1), nn.Flatten(0))
+
+    spec = ModelSpec(
+        module=types.SimpleNamespace(),
+        model_fn=model_fn,
+        loss_fn=lambda o, y: torch.nn.functional
+        .binary_cross_entropy_with_logits(o.float(), y.float()),
+        optimizer_fn=lambda m=None: ("sgd", "learning_rate=0.05"),
+    )
+    tm, ev, servicer, server = start_master(spec, reader,
+                                            records_per_task=128)
+    ps_servers, ps_addrs = start_ps(1, opt_args="learning_rate=0.05")
+    try:
+        mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=0)
+        trainer = ParameterServerTrainer(spec, PSClient(ps_addrs),
+                                         device="cpu")
+        losses = []
+        orig = trainer.train_minibatch
+
+        def recording(batch):
+            loss, v = orig(batch)
+            losses.append(float(loss))
+            return loss, v
+
+        trainer.train_minibatch = recording
+        worker = Worker(0, mc, trainer, data_reader=reader, spec=spec,
+                        minibatch_size=32)
+        worker.run()
+        first = sum(losses[:5]) / 5
+        last = sum(losses[-5:]) / 5
+        assert last < 0.5 * first, (first, last)
+    finally:
+        server.stop(0)
+        for ps in ps_servers:
+            ps.server.stop(0)
