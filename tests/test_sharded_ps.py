@@ -79,3 +79,28 @@ def test_sharded_ps_two_ranks():
             assert p.exitcode == 0
         # both ranks observed identical rows (single source of truth)
         assert torch.allclose(results[0], results[1])
+
+
+def test_sharded_ps_three_ranks():
+    """Odd world size: uneven id%3 shard populations and zero-size
+    all_to_all splits must work (SCALE runs use N in {2,4,8}; an odd
+    world is the stronger divisibility probe)."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        procs = [
+            ctx.Process(target=_worker, args=(r, 3, port, results))
+            for r in range(3)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(240)
+            assert p.exitcode == 0
+        assert torch.allclose(results[0], results[1])
+        assert torch.allclose(results[0], results[2])
