@@ -46,14 +46,15 @@ class DCN(nn.Module):
         self.head = nn.Linear(dim + dims[-1], 1)
 
     def forward(self, ids: torch.Tensor) -> torch.Tensor:
-        x0 = self.embedding(ids).flatten(1)  # [B, F*d] f32
+        # follow the module's runtime dtype (f32 on CPU, bf16 when the
+        # trainer casts the model): the PS returns f32 rows, so one cast
+        # at entry keeps cross/head math dtype-consistent
+        dtype = self.head.weight.dtype
+        x0 = self.embedding(ids).flatten(1).to(dtype)  # [B, F*d]
         x = x0
         for layer in self.cross:
             x = layer(x0, x)
-        deep_in = x0
-        if deep_in.device.type == "cuda":
-            deep_in = deep_in.to(torch.bfloat16)
-        d = self.deep(deep_in).float()
+        d = self.deep(x0).to(dtype)  # FusedDense runs bf16 on GPU
         return self.head(torch.cat([x, d], dim=1)).squeeze(-1)
 
 
