@@ -161,3 +161,42 @@ def test_worker_priority_fraction():
     assert mgr._worker_priority(2) == "low"
     mgr._priority = "high"
     assert mgr._worker_priority(3) == "high"
+
+
+def test_job_failure_callback_on_ps_death():
+    """PS death under PS strategy must fail the job (reference
+    TFV1PSStrategyTrainLoopMonitorCallback); worker deaths must not."""
+    from elasticdl_amd.master.pod_event_callbacks import JobFailureCallback
+
+    class FakeMaster:
+        stopped = None
+
+        def request_stop(self, success=True):
+            self.stopped = success
+
+    m = FakeMaster()
+    cb = JobFailureCallback(m)
+    cb.on_pod_failed(NS(type="worker", name="w0", id=0))
+    assert m.stopped is None
+    cb.on_pod_failed(NS(type="ps", name="ps-0", id=0))
+    assert m.stopped is False
+
+
+def test_k8s_pod_manager_invokes_job_failure_on_ps(monkeypatch):
+    """Wired end to end: a Failed PS pod event stops the master."""
+    mgr, fake = make_manager(num_workers=1, num_ps=1)
+
+    class FakeMaster:
+        stopped = None
+
+        def request_stop(self, success=True):
+            self.stopped = success
+
+    from elasticdl_amd.master.pod_event_callbacks import JobFailureCallback
+
+    master = FakeMaster()
+    mgr.add_pod_event_callback(JobFailureCallback(master))
+    name = "elasticdl-j-ps-0"
+    mgr._event_cb(pod_event(name, "ps", 0, "Running", PodEventType.ADDED))
+    mgr._event_cb(pod_event(name, "ps", 0, "Failed"))
+    assert master.stopped is False
