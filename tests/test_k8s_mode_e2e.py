@@ -21,6 +21,8 @@ import threading
 import time
 from types import SimpleNamespace as NS
 
+import pytest
+
 from elasticdl_amd.common.args import parse_master_args
 from elasticdl_amd.master.k8s_client import (
     ELASTICDL_REPLICA_INDEX_KEY,
@@ -75,6 +77,10 @@ class FakeCluster:
     def read_namespaced_service(self, name, ns):
         with self.lock:
             return self.services[name]
+
+    def patch_namespaced_service(self, name, ns, svc):
+        with self.lock:
+            self.services[name] = svc
 
     # -- watch
     def stream_pod_events(self, ns, label_selector=""):
@@ -205,3 +211,28 @@ def test_k8s_mode_task_timeout_kills_pod(monkeypatch):
     master.pod_manager.start_workers()
     master.pod_manager.kill_worker(1)
     assert fake.deleted == ["elasticdl-jobx-worker-1"]
+
+
+def test_service_spec_selector_index_patching():
+    """A relaunched replica can be patched behind the ORIGINAL service
+    name (reference patch_worker_service): the spec keeps the original
+    name but selects the replacement's index."""
+    fake = FakeCluster()
+    client = Client("prod", "jobx", "img", core_api=fake)
+    svc = client.build_service_spec(pod_type="worker", index=0, port=3333,
+                                    selector_index=2)
+    assert svc.metadata.name == "elasticdl-jobx-worker-0"
+    assert svc.spec.selector["elasticdl-replica-index"] == "2"
+    assert svc.spec.ports[0].port == 3333
+    assert client.create_service(svc)
+    assert client.patch_service(svc.metadata.name, svc)
+    assert fake.services["elasticdl-jobx-worker-0"] is svc
+
+
+def test_offline_spec_types_reject_unknown_fields():
+    from elasticdl_amd.master import k8s_types as t
+
+    with pytest.raises(TypeError):
+        t.V1ObjectMeta(name="x", nope=1)
+    meta = t.V1ObjectMeta(name="x")
+    assert "name='x'" in repr(meta)
