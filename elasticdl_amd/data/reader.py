@@ -307,3 +307,25 @@ def create_data_reader(data_origin: str, records_per_shard: int = 0,
             return RecordIOReader(data_origin, records_per_shard)
         return TextReader(data_origin, records_per_shard, **kwargs)
     raise ValueError(f"cannot infer reader for {data_origin!r}")
+
+
+def synthetic_reader_from_spec(spec, data_origin: str,
+                               records_per_shard: int = 0):
+    """``synthetic:<n>`` for any zoo module exporting ``synthetic_batch``:
+    record i is sample 0 of ``synthetic_batch(batch_size=1, seed=i)``, so
+    every worker regenerates identical records from the shard indices and
+    the stock collate path re-batches them. Returns None when the module
+    has no synthetic_batch (caller falls through to the file factory)."""
+    if not data_origin.startswith("synthetic:"):
+        return None
+    sb = getattr(spec.module, "synthetic_batch", None)
+    if sb is None:
+        return None
+    size = int(data_origin.split(":", 1)[1])
+
+    def sample(i: int):
+        batch = sb(batch_size=1, seed=i)
+        return tuple(t[0] if getattr(t, "ndim", 0) > 0 else t for t in batch)
+
+    return SyntheticReader(size, sample, records_per_shard,
+                           name=f"{spec.module.__name__}-synthetic")

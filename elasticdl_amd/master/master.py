@@ -46,7 +46,8 @@ class Master:
         self.stopped = threading.Event()
         self.exit_code = 0
 
-        spec = get_model_spec(args.model_def, parse_model_params(args.model_params))
+        spec = get_model_spec(args.model_def, parse_model_params(args.model_params),
+                              model_zoo=getattr(args, "model_zoo", ""))
         self.spec = spec
 
         # ---- data shards
@@ -56,9 +57,13 @@ class Master:
             reader = spec.data_reader_fn(args.training_data)
             training_shards = reader.create_shards()
         elif args.training_data:
-            from elasticdl_amd.data.reader import create_data_reader
+            from elasticdl_amd.data.reader import (
+                create_data_reader,
+                synthetic_reader_from_spec,
+            )
 
-            reader = create_data_reader(args.training_data)
+            reader = synthetic_reader_from_spec(spec, args.training_data) \
+                or create_data_reader(args.training_data)
             training_shards = reader.create_shards()
         if args.validation_data:
             from elasticdl_amd.data.reader import create_data_reader
@@ -153,6 +158,7 @@ class Master:
             "--master_addr", self.master_addr,
             "--worker_id", str(worker_id),
             "--model_def", a.model_def,
+            "--model_zoo", a.model_zoo,
             "--model_params", a.model_params,
             "--distribution_strategy", a.distribution_strategy,
             "--minibatch_size", str(a.minibatch_size),

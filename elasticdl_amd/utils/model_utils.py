@@ -28,16 +28,39 @@ _BUILTIN = {
 }
 
 
-def load_module(model_def: str):
+def _load_file(path: str):
+    spec = importlib.util.spec_from_file_location(
+        os.path.splitext(os.path.basename(path))[0], path
+    )
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    return mod
+
+
+def load_module(model_def: str, model_zoo: str = ""):
+    """Resolve a model definition to a module.
+
+    Order (reference: common/model_utils.py:27-60, where model_def is a
+    path inside the --model_zoo directory): built-in short name; a file
+    inside ``model_zoo`` (``<zoo>/<model_def>`` or ``<zoo>/<model_def>.py``,
+    dots treated as path separators so ``pkg.model`` finds
+    ``<zoo>/pkg/model.py``); a standalone file path; a dotted module on
+    sys.path (with ``model_zoo`` prepended when given).
+    """
     if model_def in _BUILTIN:
         return importlib.import_module(_BUILTIN[model_def])
+    if model_zoo:
+        rel = model_def if model_def.endswith(".py") \
+            else model_def.replace(".", os.sep) + ".py"
+        for cand in (os.path.join(model_zoo, model_def),
+                     os.path.join(model_zoo, rel)):
+            if os.path.isfile(cand):
+                return _load_file(cand)
+        import sys
+        if model_zoo not in sys.path:
+            sys.path.insert(0, model_zoo)
     if os.path.isfile(model_def):
-        spec = importlib.util.spec_from_file_location(
-            os.path.splitext(os.path.basename(model_def))[0], model_def
-        )
-        mod = importlib.util.module_from_spec(spec)
-        spec.loader.exec_module(mod)
-        return mod
+        return _load_file(model_def)
     return importlib.import_module(model_def)
 
 
@@ -57,8 +80,9 @@ class ModelSpec:
         return self.model_fn(**self.params)
 
 
-def get_model_spec(model_def: str, model_params: Optional[dict] = None) -> ModelSpec:
-    mod = load_module(model_def)
+def get_model_spec(model_def: str, model_params: Optional[dict] = None,
+                   model_zoo: str = "") -> ModelSpec:
+    mod = load_module(model_def, model_zoo)
 
     def need(name):
         fn = getattr(mod, name, None)
