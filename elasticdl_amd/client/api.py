@@ -15,7 +15,7 @@ import subprocess
 import sys
 from typing import List
 
-from elasticdl_amd.common.args import build_arguments_from_parsed_result
+from elasticdl_amd.common.args import parse_envs, build_arguments_from_parsed_result
 from elasticdl_amd.common.log_utils import default_logger as logger
 
 _MASTER_ARG_KEYS = {
@@ -81,7 +81,7 @@ def _create_master_pod(args) -> int:
         resource_requests=args.master_resource_request,
         resource_limits=args.master_resource_limit,
         priority_class=args.master_pod_priority,
-        envs={},
+        envs=parse_envs(getattr(args, "envs", "")),
         volumes=args.volume,
         image_pull_policy=args.image_pull_policy,
         restart_policy=args.restart_policy,

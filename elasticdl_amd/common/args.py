@@ -92,6 +92,21 @@ def parse_worker_args(argv=None) -> argparse.Namespace:
     return p.parse_args(argv)
 
 
+def parse_envs(s: str) -> dict:
+    """--envs 'K=V,K2=V2' -> dict (reference: extra container env vars
+    forwarded to every pod the master creates)."""
+    out = {}
+    for part in (s or "").split(","):
+        part = part.strip()
+        if not part:
+            continue
+        if "=" not in part:
+            raise ValueError(f"--envs entry {part!r} is not K=V")
+        k, v = part.split("=", 1)
+        out[k.strip()] = v.strip()
+    return out
+
+
 def parse_model_params(s: str) -> dict:
     out = {}
     for part in (s or "").split(";"):

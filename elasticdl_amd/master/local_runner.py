@@ -38,6 +38,7 @@ class LocalProcessManager:
         num_ps: int = 0,
         relaunch_on_worker_failure: int = 0,
         log_dir: str = "",
+        user_envs: Optional[Dict[str, str]] = None,
     ):
         self.master_addr = master_addr
         self.worker_command = worker_command
@@ -46,6 +47,7 @@ class LocalProcessManager:
         self.num_ps = num_ps
         self.relaunch_on_worker_failure = relaunch_on_worker_failure
         self.log_dir = log_dir
+        self.user_envs = dict(user_envs or {})
         self.procs: Dict[str, LocalProcess] = {}
         self._lock = threading.Lock()
         self._next_worker_id = 0
@@ -92,6 +94,7 @@ class LocalProcessManager:
             filter(None, [repo_root, env.get("PYTHONPATH", "")])
         )
         env[WorkerEnv.MASTER_ADDR] = self.master_addr
+        env.update(self.user_envs)
         env.update(extra_env)
         stdout = None
         if self.log_dir:

@@ -14,7 +14,7 @@ import threading
 import time
 from typing import List, Optional
 
-from elasticdl_amd.common.args import parse_model_params
+from elasticdl_amd.common.args import parse_envs, parse_model_params
 from elasticdl_amd.common.constants import DistributionStrategy
 from elasticdl_amd.common.log_utils import default_logger as logger
 from elasticdl_amd.common.rpc import start_server
@@ -218,6 +218,7 @@ class Master:
                 a, "relaunch_on_worker_failure", 3
             ),
             log_dir=os.path.join(a.checkpoint_dir or "/tmp/edl", "logs"),
+            user_envs=parse_envs(getattr(a, "envs", "")),
         )
         mgr.ps_addrs = self.ps_addrs
         mgr.add_pod_event_callback(TaskRescheduleCallback(self.task_manager))

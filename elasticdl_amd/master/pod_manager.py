@@ -126,11 +126,14 @@ class PodManager:
         owner = self.k8s.get_pod(self.k8s.get_master_pod_name())
         from elasticdl_amd.common.constants import WorkerEnv
 
-        envs = {
+        from elasticdl_amd.common.args import parse_envs
+
+        envs = parse_envs(getattr(self.args, "envs", ""))
+        envs.update({
             WorkerEnv.MASTER_ADDR: self.master.master_addr,
             WorkerEnv.WORKER_ID: str(index),
             WorkerEnv.WORKER_NUM: str(self.args.num_workers),
-        }
+        })
         pod = self.k8s.build_pod_spec(
             pod_name=name,
             pod_type=pod_type,
