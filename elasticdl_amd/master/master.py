@@ -51,34 +51,32 @@ class Master:
         self.spec = spec
 
         # ---- data shards
-        reader = None
-        training_shards = evaluation_shards = None
-        if spec.data_reader_fn is not None:
-            reader = spec.data_reader_fn(args.training_data)
-            training_shards = reader.create_shards()
-        elif args.training_data:
+        def _make_reader(origin: str):
+            """Same resolution chain for train/validation/prediction data:
+            zoo custom_data_reader > generic synthetic:<n> > file factory."""
+            if spec.data_reader_fn is not None:
+                return spec.data_reader_fn(origin)
             from elasticdl_amd.data.reader import (
                 create_data_reader,
                 synthetic_reader_from_spec,
             )
 
-            reader = synthetic_reader_from_spec(spec, args.training_data) \
-                or create_data_reader(args.training_data)
+            return synthetic_reader_from_spec(spec, origin) \
+                or create_data_reader(origin)
+
+        reader = None
+        training_shards = evaluation_shards = None
+        if args.training_data or spec.data_reader_fn is not None:
+            reader = _make_reader(args.training_data)
             training_shards = reader.create_shards()
         if args.validation_data:
-            from elasticdl_amd.data.reader import create_data_reader
-
-            evaluation_shards = create_data_reader(
-                args.validation_data
-            ).create_shards()
+            evaluation_shards = _make_reader(
+                args.validation_data).create_shards()
 
         prediction_shards = None
         if args.prediction_data:
-            from elasticdl_amd.data.reader import create_data_reader
-
-            prediction_shards = create_data_reader(
-                args.prediction_data
-            ).create_shards()
+            prediction_shards = _make_reader(
+                args.prediction_data).create_shards()
 
         # job type derivation (reference: elasticdl_job_service.py:32-54)
         self.job_type = args.job_type or (
@@ -165,6 +163,7 @@ class Master:
             "--get_model_steps", str(a.get_model_steps),
             "--training_data", a.training_data,
             "--validation_data", a.validation_data,
+            "--prediction_data", a.prediction_data,
             "--device", a.device,
             "--log_loss_steps", str(a.log_loss_steps),
         ]

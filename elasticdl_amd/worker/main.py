@@ -31,25 +31,29 @@ def build_worker(args) -> Worker:
                           model_zoo=getattr(args, "model_zoo", ""))
     mc = MasterClient(master_addr, worker_id)
 
-    reader = None
-    eval_reader = None
-    if spec.data_reader_fn is not None:
-        reader = spec.data_reader_fn(args.training_data)
-        if args.validation_data:
-            eval_reader = spec.data_reader_fn(args.validation_data)
-    else:
+    def _make_reader(origin: str):
+        # same chain as the master: zoo custom_data_reader >
+        # generic synthetic:<n> > file factory
+        if spec.data_reader_fn is not None:
+            return spec.data_reader_fn(origin)
         from elasticdl_amd.data.reader import (
             create_data_reader,
             synthetic_reader_from_spec,
         )
 
-        if args.training_data:
-            reader = synthetic_reader_from_spec(spec, args.training_data) \
-                or create_data_reader(args.training_data)
-        if args.validation_data:
-            eval_reader = synthetic_reader_from_spec(
-                spec, args.validation_data) \
-                or create_data_reader(args.validation_data)
+        return synthetic_reader_from_spec(spec, origin) \
+            or create_data_reader(origin)
+
+    reader = None
+    eval_reader = None
+    if args.training_data or spec.data_reader_fn is not None:
+        reader = _make_reader(args.training_data)
+    if args.validation_data:
+        eval_reader = _make_reader(args.validation_data)
+    elif args.prediction_data:
+        # predict jobs shard --prediction_data; those tasks read through
+        # the eval reader slot (worker.py routes PREDICTION there)
+        eval_reader = _make_reader(args.prediction_data)
 
     if args.distribution_strategy == DistributionStrategy.PARAMETER_SERVER:
         from elasticdl_amd.worker.ps_client import PSClient

@@ -115,3 +115,38 @@ def test_local_job_with_custom_zoo_dir(zoo, tmp_path):
                        cwd=REPO, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
     assert os.path.exists(export)
+
+
+@pytest.mark.timeout(300)
+def test_predict_only_job_subprocess(zoo, tmp_path):
+    """Predict-only job through real processes: --prediction_data is
+    sharded by the master, forwarded to the worker command line, and the
+    worker reads those shards through its reader chain (this used to
+    leave the worker with no reader at all)."""
+    outfile = str(tmp_path / "preds.txt")
+    (tmp_path / "zoo2").mkdir()
+    mod = ZOO_MODULE + textwrap.dedent(
+        """
+        def process_predictions(outputs):
+            with open(%r, "a") as f:
+                f.write("%%d\\n" %% outputs.shape[0])
+        """ % outfile
+    )
+    (tmp_path / "zoo2" / "linear.py").write_text(mod)
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", str(tmp_path / "zoo2"),
+        "--model_def", "linear",
+        "--job_type", "predict",
+        "--num_workers", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--prediction_data", "synthetic:64",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    with open(outfile) as f:
+        assert sum(int(x) for x in f.read().split()) == 64
