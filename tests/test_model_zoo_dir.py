@@ -150,3 +150,25 @@ def test_predict_only_job_subprocess(zoo, tmp_path):
     assert r.returncode == 0, r.stderr[-3000:]
     with open(outfile) as f:
         assert sum(int(x) for x in f.read().split()) == 64
+
+
+@pytest.mark.timeout(300)
+def test_evaluate_only_job_subprocess(zoo, tmp_path):
+    """Evaluate-only job through real processes: checkpoint restore +
+    eval tasks from --validation_data, no training."""
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", zoo,
+        "--model_def", "linear",
+        "--job_type", "evaluate",
+        "--num_workers", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--validation_data", "synthetic:64",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert "accuracy" in (r.stdout + r.stderr)
