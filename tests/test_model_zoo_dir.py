@@ -172,3 +172,81 @@ def test_evaluate_only_job_subprocess(zoo, tmp_path):
                        cwd=REPO, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
     assert "accuracy" in (r.stdout + r.stderr)
+
+
+@pytest.mark.timeout(300)
+def test_checkpoint_restore_full_stack(tmp_path):
+    """--checkpoint_dir_for_init through real processes: a hand-written
+    checkpoint with a perfect separator is restored by the PS daemon,
+    the worker adopts the PS params (not its own random init), and an
+    evaluate-only job reports accuracy 1.0."""
+    from elasticdl_amd.utils.save_utils import CheckpointSaver
+
+    zoo2 = tmp_path / "zoo"
+    zoo2.mkdir()
+    (zoo2 / "sep.py").write_text(textwrap.dedent(
+        """
+        import torch
+        import torch.nn as nn
+
+        def custom_model(**kw):
+            return nn.Sequential(nn.Linear(8, 2))
+
+        def loss(outputs, labels):
+            return nn.functional.cross_entropy(outputs, labels.long())
+
+        def optimizer(model=None):
+            return ("sgd", "learning_rate=0.01")
+
+        def eval_metrics_fn():
+            return {"accuracy":
+                    lambda out, lab: (out.argmax(1) == lab).float().mean()}
+
+        def feed(batch, device, dtype=None):
+            x, y = batch
+            x = x.to(device)
+            if dtype is not None:
+                x = x.to(dtype)
+            return x, y.to(device)
+
+        def synthetic_batch(batch_size=32, seed=None):
+            g = torch.Generator().manual_seed(seed) \\
+                if seed is not None else None
+            x = torch.randn(batch_size, 8, generator=g)
+            y = (x[:, 0] > 0).long()
+            return x, y
+        """
+    ))
+    # perfect separator on y = (x0 > 0): logits = [-10*x0, +10*x0]
+    w = torch.zeros(2, 8)
+    w[0, 0], w[1, 0] = -10.0, 10.0
+    ckpt = tmp_path / "ckpt"
+    CheckpointSaver(str(ckpt)).save_shard(
+        version=7,
+        state={"version": 7,
+               "dense": {"0.weight": w, "0.bias": torch.zeros(2)},
+               "embedding_tables": {}, "embedding_infos": []},
+        shard_id=0, num_shards=1,
+    )
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", str(zoo2),
+        "--model_def", "sep",
+        "--job_type", "evaluate",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--validation_data", "synthetic:64",
+        "--checkpoint_dir_for_init", str(ckpt),
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    out = r.stdout + r.stderr
+    import re
+    m = re.search(r"accuracy[^0-9]*([01]\.\d+)", out)
+    assert m, out[-2000:]
+    assert float(m.group(1)) == 1.0, m.group(1)
