@@ -250,3 +250,30 @@ def test_checkpoint_restore_full_stack(tmp_path):
     m = re.search(r"accuracy[^0-9]*([01]\.\d+)", out)
     assert m, out[-2000:]
     assert float(m.group(1)) == 1.0, m.group(1)
+
+
+@pytest.mark.timeout(300)
+def test_train_with_interleaved_eval_subprocess(zoo, tmp_path):
+    """Training job with --evaluation_steps: version-triggered eval tasks
+    interleave with training through real processes and the master logs
+    aggregated metrics."""
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", zoo,
+        "--model_def", "linear",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--num_epochs", "2",
+        "--shuffle", "true",
+        "--training_data", "synthetic:128",
+        "--validation_data", "synthetic:32",
+        "--evaluation_steps", "3",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert "accuracy" in (r.stdout + r.stderr)
