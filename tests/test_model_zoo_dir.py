@@ -277,3 +277,39 @@ def test_train_with_interleaved_eval_subprocess(zoo, tmp_path):
                        cwd=REPO, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
     assert "accuracy" in (r.stdout + r.stderr)
+
+
+@pytest.mark.timeout(300)
+def test_checkpoint_write_then_resume_subprocess(zoo, tmp_path):
+    """Train to --max_step with periodic PS checkpoints, then resume
+    from them: PS params restore, master fast-forwards completed steps,
+    and the resumed job trains only the remainder."""
+    ckpt = str(tmp_path / "ckpt")
+    base = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", zoo,
+        "--model_def", "linear",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "1",
+        "--num_epochs", "8",
+        "--training_data", "synthetic:128",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    env = dict(os.environ, PYTHONPATH=REPO)
+    r = subprocess.run(base + ["--max_step", "4",
+                               "--checkpoint_dir", ckpt,
+                               "--checkpoint_steps", "2"],
+                       env=env, cwd=REPO, capture_output=True, text=True,
+                       timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    versions = sorted(d for d in os.listdir(ckpt) if d.startswith("version-"))
+    assert versions, "no checkpoints written"
+
+    r = subprocess.run(base + ["--max_step", "6",
+                               "--checkpoint_dir_for_init", ckpt],
+                       env=env, cwd=REPO, capture_output=True, text=True,
+                       timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
