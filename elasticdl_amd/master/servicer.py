@@ -92,6 +92,20 @@ class MasterServicer:
         ):
             if self._task_manager.pending_evaluation_tasks == 0:
                 self._evaluation_service.complete_task()
+        if (
+            in_doing
+            and not err
+            and task is not None
+            and task.type == TaskType.TRAINING
+            and self._evaluation_service is not None
+            and self._rendezvous_server is not None
+        ):
+            # AllReduce jobs have no PS reporting model versions; trigger
+            # version-keyed evaluation from completed training steps
+            # (version == global step in collective mode).
+            self._evaluation_service.add_evaluation_task_if_needed(
+                self._task_manager.completed_steps
+            )
         return {}
 
     def report_training_params(self, req: dict) -> dict:

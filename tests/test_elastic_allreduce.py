@@ -72,3 +72,19 @@ def test_allreduce_worker_killed_job_survives():
             pass  # finished already; elasticity path still validated below
         out, _ = p.communicate(timeout=650)
         assert p.returncode == 0, out[-6000:]
+
+
+@pytest.mark.timeout(720)
+def test_allreduce_with_interleaved_eval():
+    """AllReduce strategy with --evaluation_steps: eval tasks interleave
+    with the collective training loop (each worker evaluates on its own
+    eval shards; master aggregates metrics)."""
+    p = run_master([
+        "--num_workers", "2",
+        "--training_data", "synthetic:256",
+        "--validation_data", "synthetic:64",
+        "--evaluation_steps", "4",
+    ])
+    out, _ = p.communicate(timeout=600)
+    assert p.returncode == 0, out[-3000:]
+    assert "accuracy" in out
