@@ -200,6 +200,13 @@ class LocalProcessManager:
                 if p.type == "worker" and p.status == PodStatus.RUNNING
             )
 
+    def get_alive_worker_ids(self):
+        with self._lock:
+            return sorted(
+                p.id for p in self.procs.values()
+                if p.type == "worker" and p.status == PodStatus.RUNNING
+            )
+
     def all_workers_exited(self) -> bool:
         with self._lock:
             workers = [p for p in self.procs.values() if p.type == "worker"]

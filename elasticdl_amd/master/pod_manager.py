@@ -220,6 +220,15 @@ class PodManager:
                                  PodStatus.INITIAL)
             )
 
+    def get_alive_worker_ids(self):
+        with self._lock:
+            return sorted(
+                p.id for p in self.pods.values()
+                if p.type == "worker"
+                and p.status in (PodStatus.PENDING, PodStatus.RUNNING,
+                                 PodStatus.INITIAL)
+            )
+
     def all_workers_exited(self) -> bool:
         with self._lock:
             workers = [p for p in self.pods.values() if p.type == "worker"]

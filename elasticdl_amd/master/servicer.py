@@ -66,13 +66,18 @@ class MasterServicer:
         worker_id = req.get("worker_id", -1)
         task = self._task_manager.get(worker_id)
         if task.type == TaskType.WAIT and self._rendezvous_server is not None:
-            # AllReduce: surplus workers exit instead of waiting
+            # AllReduce: surplus workers exit instead of waiting, but ONE
+            # worker (the lowest alive id) must stay to drain the tail —
+            # e.g. the train-end export task emitted once the last
+            # in-flight training task resolves (reference servicer.py:
+            # 111-125: only the LAST live worker gets WAIT).
             alive = (
-                self._pod_manager.get_alive_worker_num()
+                self._pod_manager.get_alive_worker_ids()
                 if self._pod_manager is not None
-                else 1
+                and hasattr(self._pod_manager, "get_alive_worker_ids")
+                else []
             )
-            if alive > 1:
+            if len(alive) > 1 and worker_id != alive[0]:
                 task = Task(task_id=0, shard=None, type=TaskType.NONE)
         return task.to_wire()
 

@@ -88,3 +88,18 @@ def test_allreduce_with_interleaved_eval():
     out, _ = p.communicate(timeout=600)
     assert p.returncode == 0, out[-3000:]
     assert "accuracy" in out
+
+
+@pytest.mark.timeout(720)
+def test_allreduce_train_end_export(tmp_path):
+    """--output in AllReduce mode: the train-end callback task routes to
+    one worker which exports the final model."""
+    export = str(tmp_path / "model.pt")
+    p = run_master([
+        "--num_workers", "2",
+        "--training_data", "synthetic:128",
+        "--output", export,
+    ])
+    out, _ = p.communicate(timeout=600)
+    assert p.returncode == 0, out[-3000:]
+    assert os.path.exists(export), out[-2000:]

@@ -11,10 +11,22 @@ from elasticdl_amd.master.task_manager import TaskManager
 
 class FakePodManager:
     def __init__(self, alive):
-        self.alive = alive
+        # accept a count (ids 0..n-1) or an explicit id list
+        self.ids = list(range(alive)) if isinstance(alive, int) else list(alive)
+
+    @property
+    def alive(self):
+        return len(self.ids)
+
+    @alive.setter
+    def alive(self, n):
+        self.ids = list(range(n))
 
     def get_alive_worker_num(self):
-        return self.alive
+        return len(self.ids)
+
+    def get_alive_worker_ids(self):
+        return sorted(self.ids)
 
 
 def make_rdzv():
@@ -32,8 +44,11 @@ def test_allreduce_surplus_workers_get_none_last_gets_wait():
     servicer = MasterServicer(tm, rendezvous_server=make_rdzv(), pod_manager=pm)
     t = servicer.get_task({"worker_id": 0})
     assert t["type"] == TaskType.TRAINING
-    # tasks in flight, 2 alive -> surplus worker told to exit
+    # tasks in flight, 2 alive -> the surplus (non-lowest-id) worker is
+    # told to exit; the designated lowest-id worker WAITS so it can
+    # drain tail tasks (train-end export) later
     assert servicer.get_task({"worker_id": 1})["type"] == TaskType.NONE
+    assert servicer.get_task({"worker_id": 0})["type"] == TaskType.WAIT
     pm.alive = 1
     assert servicer.get_task({"worker_id": 0})["type"] == TaskType.WAIT
 
