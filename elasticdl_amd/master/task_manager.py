@@ -204,6 +204,7 @@ class TaskManager:
             )
         )
         self._train_end_callback_emitted = True
+        logger.info("TaskManager: emitted train-end callback task")
 
     # ---------------------------------------------------------------- workers
     def register_task_timeout_callback(self, fn: Callable[[int], None]) -> None:

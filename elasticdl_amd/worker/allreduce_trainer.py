@@ -208,9 +208,13 @@ class AllReduceTrainer(Trainer):
         return self._version
 
     def export_model(self, path: str) -> None:
-        if self.comm.rank <= 0:
-            torch.save(self.model.state_dict(), path)
-            logger.info("Exported model to %s", path)
+        # Called from the ONE worker the master handed the train-end task
+        # (post-allreduce the model is identical on every rank, and the
+        # rest of the world may already have exited) — never rank-gate
+        # here or the export silently vanishes when the task lands on a
+        # non-zero rank.
+        torch.save(self.model.state_dict(), path)
+        logger.info("Exported model to %s", path)
 
     def on_training_end(self) -> None:
         from elasticdl_amd.master.servicer import TrainingLoopStatus
