@@ -131,9 +131,21 @@ class Worker:
                 err = e
                 if attempt < 2:
                     logger.warning("minibatch failed (%s); retrying", e)
+                import grpc
+
+                if attempt >= 2 and not isinstance(
+                    e, (grpc.RpcError, ConnectionError, OSError,
+                        TimeoutError)
+                ):
+                    # the 64-retry budget exists for transient PS/RPC
+                    # unavailability (reference worker.py:39); a
+                    # deterministic model-side error (shape mismatch,
+                    # dtype, bad zoo code) will fail all 64 times — bail
+                    # after 3 so the job fails in seconds, not minutes
+                    break
                 time.sleep(min(0.1 * (attempt + 1), 2.0))
         raise RuntimeError(
-            f"minibatch failed after {MAX_MINIBATCH_RETRY_NUM} retries"
+            f"minibatch failed ({err}); retries exhausted"
         ) from err
 
     def _run_training_task(self, task: Task) -> None:

@@ -313,3 +313,27 @@ def test_checkpoint_write_then_resume_subprocess(zoo, tmp_path):
                        env=env, cwd=REPO, capture_output=True, text=True,
                        timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
+
+
+@pytest.mark.timeout(300)
+def test_two_ps_two_workers_subprocess(tmp_path):
+    """Sharded PS through real processes: dense params split by name
+    hash and embedding rows by id-mod across 2 PS daemons, 2 workers
+    pushing concurrently; the job trains and exports."""
+    export = str(tmp_path / "model.pt")
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "deepfm",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "2", "--num_ps_pods", "2",
+        "--minibatch_size", "32",
+        "--num_minibatches_per_task", "2",
+        "--training_data", "synthetic:256",
+        "--device", "cpu",
+        "--output", export,
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert os.path.exists(export)

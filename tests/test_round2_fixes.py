@@ -291,3 +291,29 @@ def test_parse_model_params_literals_only():
     # expressions must NOT be evaluated
     out = parse_model_params("x=(1).__class__")
     assert out["x"] == "(1).__class__"
+
+
+def test_deterministic_minibatch_error_fails_fast(monkeypatch):
+    """Model-side (non-RPC) errors must not burn the full 64-retry
+    budget: a shape mismatch fails the task after ~3 attempts."""
+    import time as _time
+
+    from elasticdl_amd.worker import worker as worker_mod
+
+    class BadTrainer:
+        calls = 0
+
+        def train_minibatch(self, batch):
+            BadTrainer.calls += 1
+            raise RuntimeError("mat1 and mat2 shapes cannot be multiplied")
+
+    w = worker_mod.Worker.__new__(worker_mod.Worker)
+    w.trainer = BadTrainer()
+    w._step = 0
+    w.log_loss_steps = 100
+    monkeypatch.setattr(_time, "sleep", lambda s: None)
+    import pytest as _pytest
+
+    with _pytest.raises(RuntimeError):
+        w._process_minibatch(batch=None, train=True)
+    assert BadTrainer.calls <= 4
