@@ -56,9 +56,16 @@ def feed(batch, device, dtype=None):
 
 
 def custom_data_reader(data_origin: str = ""):
-    """Synthetic MNIST-shaped reader (no network -> no real dataset);
-    data_origin may be 'synthetic:<size>'."""
-    from elasticdl_amd.data.reader import SyntheticReader
+    """'synthetic:<size>' generates MNIST-shaped records in-process; a
+    real path (e.g. a RecordIO dir from recordio_gen.gen_mnist_recordio)
+    goes through the normal reader factory and collate_fn decodes the
+    encoded records."""
+    import os
+
+    from elasticdl_amd.data.reader import SyntheticReader, create_data_reader
+
+    if data_origin and os.path.exists(data_origin):
+        return create_data_reader(data_origin)
 
     size = 640
     if data_origin.startswith("synthetic:"):
@@ -72,6 +79,19 @@ def custom_data_reader(data_origin: str = ""):
         )
 
     return SyntheticReader(size, sample, name="mnist-synthetic")
+
+
+def collate_fn(records):
+    """Tuple records (synthetic) stack with the default collate; bytes
+    records (RecordIO) decode via recordio_gen's codec."""
+    if records and isinstance(records[0], (bytes, bytearray)):
+        from elasticdl_amd.data.recordio_gen import collate_records
+
+        x, y = collate_records(records)
+        return x.unsqueeze(1), y  # [n,28,28] -> [n,1,28,28] (conv input)
+    from torch.utils.data import default_collate
+
+    return default_collate(records)
 
 
 def synthetic_batch(batch_size: int = 64, seed: int = None):

@@ -337,3 +337,32 @@ def test_two_ps_two_workers_subprocess(tmp_path):
                        cwd=REPO, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
     assert os.path.exists(export)
+
+
+@pytest.mark.timeout(300)
+def test_recordio_training_job_subprocess(tmp_path):
+    """Train mnist from RecordIO files on disk through real processes:
+    generator -> directory of chunked files -> factory sniffing ->
+    shard creation from the index -> workers decode records."""
+    from elasticdl_amd.data.recordio_gen import gen_mnist_recordio
+
+    data_dir = str(tmp_path / "mnist_rio")
+    files = gen_mnist_recordio(data_dir, n=192, records_per_file=64)
+    assert len(files) == 3
+    export = str(tmp_path / "model.pt")
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "mnist",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--training_data", data_dir,
+        "--device", "cpu",
+        "--output", export,
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert os.path.exists(export)
