@@ -366,3 +366,33 @@ def test_recordio_training_job_subprocess(tmp_path):
                        cwd=REPO, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
     assert os.path.exists(export)
+
+
+@pytest.mark.timeout(300)
+def test_csv_training_job_subprocess(tmp_path):
+    """Train iris from a CSV file on disk through real processes
+    (TextReader line-range shards, zoo collate parses rows)."""
+    import random
+
+    rng = random.Random(0)
+    csv_path = tmp_path / "iris.csv"
+    with open(csv_path, "w") as f:
+        f.write("f0,f1,f2,f3,label\n")
+        for _ in range(120):
+            c = rng.randint(0, 2)
+            f.write(",".join(f"{rng.gauss(c, 0.3):.3f}" for _ in range(4))
+                    + f",{c}\n")
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "iris",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--training_data", str(csv_path),
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
