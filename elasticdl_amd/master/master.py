@@ -285,6 +285,19 @@ class Master:
                     break
         finally:
             self.cleanup()
+        # a job whose data was entirely dropped (every task exhausted its
+        # retries) must not report success — reference counts failed
+        # records (task_manager.py:71-93); we additionally fail the job
+        # when NOTHING succeeded
+        tm = self.task_manager
+        if tm.failed_records:
+            logger.warning(
+                "Job dropped %d records after task retries (completed %d)",
+                tm.failed_records, tm.counts()["completed_records"],
+            )
+            if tm.counts()["completed_records"] == 0:
+                logger.error("All task shards failed; marking job failed")
+                self.exit_code = 1
         return self.exit_code
 
     def request_stop(self, success: bool = True) -> None:

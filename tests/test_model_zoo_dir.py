@@ -396,3 +396,37 @@ def test_csv_training_job_subprocess(tmp_path):
     r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
                        cwd=REPO, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
+
+
+@pytest.mark.timeout(300)
+def test_broken_model_fails_job_cleanly(zoo, tmp_path):
+    """A deterministically-broken zoo module must FAIL the job quickly
+    (fail-fast minibatch retries -> task retries <=3 -> all workers
+    failed -> master exit != 0), not hang for the full 64-retry budget."""
+    (tmp_path / "bad").mkdir()
+    (tmp_path / "bad" / "broken.py").write_text(
+        ZOO_MODULE.replace(
+            "return nn.functional.cross_entropy(outputs, labels.long())",
+            "raise RuntimeError('bad loss')",
+        )
+    )
+    import time
+
+    t0 = time.monotonic()
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", str(tmp_path / "bad"),
+        "--model_def", "broken",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--training_data", "synthetic:64",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    elapsed = time.monotonic() - t0
+    assert r.returncode != 0
+    assert elapsed < 120, f"failure took {elapsed:.0f}s"
