@@ -430,3 +430,33 @@ def test_broken_model_fails_job_cleanly(zoo, tmp_path):
     elapsed = time.monotonic() - t0
     assert r.returncode != 0
     assert elapsed < 120, f"failure took {elapsed:.0f}s"
+
+
+@pytest.mark.timeout(300)
+def test_worker_crash_exhausts_relaunch_budget(tmp_path):
+    """A worker that dies at startup is relaunched up to
+    --relaunch_on_worker_failure times; then the job fails (exit 1,
+    'All workers failed')."""
+    (tmp_path / "z").mkdir()
+    (tmp_path / "z" / "wcrash.py").write_text(
+        "import os\n"
+        "if os.environ.get('EDL_WORKER_ID') is not None:\n"
+        "    raise SystemExit(3)\n" + ZOO_MODULE
+    )
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", str(tmp_path / "z"),
+        "--model_def", "wcrash",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--training_data", "synthetic:64",
+        "--relaunch_on_worker_failure", "2",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 1
+    out = r.stdout + r.stderr
+    assert out.count("Relaunching failed worker") == 2
+    assert "All workers failed" in out
