@@ -460,3 +460,40 @@ def test_worker_crash_exhausts_relaunch_budget(tmp_path):
     out = r.stdout + r.stderr
     assert out.count("Relaunching failed worker") == 2
     assert "All workers failed" in out
+
+
+@pytest.mark.timeout(300)
+def test_ps_death_fails_job_fast_local():
+    """PS failure is NOT tolerated (reference pod_event_callbacks.py:
+    118-150): SIGKILL the PS daemon mid-job; the master stops the job
+    with exit 1 within seconds."""
+    import signal
+    import time
+
+    tmp = tempfile.mkdtemp()
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "mnist",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--training_data", "synthetic:20000",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--checkpoint_dir", tmp,
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    p = subprocess.Popen(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                         cwd=REPO, stdout=subprocess.PIPE,
+                         stderr=subprocess.STDOUT, text=True)
+    pid_file = os.path.join(tmp, "logs", "ps-0.pid")
+    deadline = time.monotonic() + 60
+    while time.monotonic() < deadline and not os.path.exists(pid_file):
+        time.sleep(0.2)
+    time.sleep(5)  # let the job actually start training
+    os.kill(int(open(pid_file).read()), signal.SIGKILL)
+    t0 = time.monotonic()
+    out, _ = p.communicate(timeout=60)
+    assert p.returncode == 1, out[-2000:]
+    assert time.monotonic() - t0 < 30
+    assert "died; stopping job" in out
