@@ -497,3 +497,41 @@ def test_ps_death_fails_job_fast_local():
     assert p.returncode == 1, out[-2000:]
     assert time.monotonic() - t0 < 30
     assert "died; stopping job" in out
+
+
+@pytest.mark.timeout(300)
+def test_hung_worker_killed_and_job_completes(tmp_path):
+    """Task-timeout watchdog end-to-end (reference task_manager.py:
+    592-616 + master.py:46-49): worker 0 hangs forever in its first
+    minibatch; the watchdog requeues its task and kills the pod; the
+    relaunched worker finishes the job."""
+    (tmp_path / "z").mkdir()
+    (tmp_path / "z" / "hang.py").write_text(ZOO_MODULE + textwrap.dedent(
+        """
+        import os, time as _t
+        _real_loss = loss
+
+        def loss(outputs, labels):
+            if os.environ.get("EDL_WORKER_ID") == "0":
+                _t.sleep(3600)
+            return _real_loss(outputs, labels)
+        """
+    ))
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", str(tmp_path / "z"),
+        "--model_def", "hang",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--training_data", "synthetic:64",
+        "--task_timeout_sec", "6",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    assert "considered hung" in out
