@@ -115,3 +115,47 @@ def test_to_ragged_to_sparse_round_trip(rows):
     sp = ToSparse()(padded)
     nnz_expected = sum(1 for r in rows for v in r if v != PAD)
     assert sp._nnz() == nnz_expected
+
+
+@given(
+    st.dictionaries(
+        st.sampled_from(["MIOPEN_FIND_MODE", "A_B", "X9", "EDL_FLAG"]),
+        st.text(alphabet=st.characters(
+            whitelist_categories=("Lu", "Ll", "Nd"),
+            whitelist_characters="=:/._-"), min_size=1, max_size=12),
+        max_size=4,
+    )
+)
+def test_parse_envs_roundtrip(d):
+    from elasticdl_amd.common.args import parse_envs
+
+    s = ",".join(f"{k}={v}" for k, v in d.items())
+    assert parse_envs(s) == d
+
+
+@given(
+    st.integers(1, 512), st.integers(1, 64), st.integers(1, 8),
+    st.booleans(), st.booleans(),
+)
+@settings(max_examples=50, deadline=None)
+def test_master_args_roundtrip(records, mb, nw, use_async, shuffle):
+    """CLI -> argv -> reparse preserves every master-relevant flag
+    (the 3-layer arg mirroring the reference round-trips the same way,
+    elasticdl_client/common/args.py:587-625)."""
+    from elasticdl_amd.common.args import (
+        build_arguments_from_parsed_result,
+        parse_master_args,
+    )
+
+    argv = [
+        "--training_data", f"synthetic:{records}",
+        "--minibatch_size", str(mb),
+        "--num_workers", str(nw),
+        "--use_async", str(use_async),
+        "--shuffle", str(shuffle),
+        "--model_params", "hidden=[64];p=0.5",
+        "--envs", "A=1,B=2",
+    ]
+    a1 = parse_master_args(argv)
+    a2 = parse_master_args(build_arguments_from_parsed_result(a1))
+    assert vars(a1) == vars(a2)
