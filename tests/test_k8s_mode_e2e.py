@@ -236,3 +236,44 @@ def test_offline_spec_types_reject_unknown_fields():
         t.V1ObjectMeta(name="x", nope=1)
     meta = t.V1ObjectMeta(name="x")
     assert "name='x'" in repr(meta)
+
+
+def test_k8s_allreduce_pod_death_refreshes_rendezvous(monkeypatch):
+    """AllReduce in k8s mode: a worker pod's death flows watch ->
+    state machine -> RendezvousServiceRefreshCallback -> staged removal
+    from the elastic rendezvous (reference pod_event_callbacks.py:
+    100-116)."""
+    fake = FakeCluster()
+    monkeypatch.setenv("MY_POD_IP", "10.9.9.9")
+    args = parse_master_args([
+        "--pod_manager", "k8s",
+        "--job_name", "jobar",
+        "--namespace", "prod",
+        "--image_name", "img",
+        "--model_def", "mnist",
+        "--distribution_strategy", "AllreduceStrategy",
+        "--num_workers", "2",
+        "--training_data", "synthetic:64",
+        "--device", "cpu",
+    ])
+    client = Client("prod", "jobar", "img", core_api=fake)
+    master = Master(args, k8s_client=client)
+    rz = master.rendezvous_server
+    assert rz is not None
+    mgr = master.pod_manager
+    mgr.start()
+    mgr.start_workers()
+    for name in list(fake.pods):
+        fake.set_phase(name, "Running")
+    # workers register with the rendezvous as worker-<id> (the identity
+    # MasterClient reports on report_training_loop_status START)
+    rz.add_worker("worker-0")
+    rz.add_worker("worker-1")
+    assert _wait(lambda: mgr.get_alive_worker_num() == 2)
+
+    fake.set_phase("elasticdl-jobar-worker-1", "Failed", exit_code=137)
+    assert _wait(
+        lambda: rz._next_hosts is not None
+        and "worker-1" not in rz._next_hosts
+    ), rz._next_hosts
+    fake.events.put(None)
