@@ -304,3 +304,64 @@ def test_single_worker_ps_actually_learns():
         server.stop(0)
         for ps in ps_servers:
             ps.server.stop(0)
+
+
+def test_get_model_steps_local_updates():
+    """train_with_local_model (reference worker.py:305-388): with
+    --get_model_steps 4 the worker pulls dense params every 4th step and
+    applies local-optimizer updates in between — and the 1-worker async
+    job still learns (the local updates carry the progress between
+    pulls)."""
+    spec = get_model_spec("iris")
+    import torch
+
+    def sample(i):
+        g = torch.Generator().manual_seed(i)
+        c = i % 3
+        x = torch.randn(4, generator=g) * 0.25 + float(c)
+        return [*x.tolist(), c]
+
+    from elasticdl_amd.data.reader import SyntheticReader
+
+    reader = SyntheticReader(4096, sample, name="iris-sep")
+    tm, ev, servicer, server = start_master(spec, reader,
+                                            records_per_task=512)
+    ps_servers, ps_addrs = start_ps(1, opt_args="learning_rate=0.05")
+    try:
+        mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=0)
+        client = PSClient(ps_addrs)
+        pulls = {"n": 0}
+        real_pull = client.pull_dense_parameters
+
+        def counted_pull(*a, **kw):
+            pulls["n"] += 1
+            return real_pull(*a, **kw)
+
+        client.pull_dense_parameters = counted_pull
+        trainer = ParameterServerTrainer(spec, client, device="cpu",
+                                         get_model_steps=4)
+        losses = []
+        real_train = trainer.train_minibatch
+
+        def recording_train(batch):
+            loss, v = real_train(batch)
+            losses.append(float(loss))
+            return loss, v
+
+        trainer.train_minibatch = recording_train
+        worker = Worker(0, mc, trainer, data_reader=reader, spec=spec,
+                        minibatch_size=32)
+        worker.run()
+        assert tm.finished()
+        n_steps = len(losses)
+        assert n_steps == 128  # 4096 / 32
+        # pulls gated to every 4th step (one extra initial _get_model)
+        assert pulls["n"] <= n_steps // 4 + 2, pulls["n"]
+        # it learned: tail loss well below the first-step loss
+        head = sum(losses[:4]) / 4
+        tail = sum(losses[-4:]) / 4
+        assert tail < head * 0.6, (head, tail)
+    finally:
+        server.stop(0)
+        for ps in ps_servers:
+            ps.server.stop(0)
