@@ -32,6 +32,7 @@ _MASTER_ARG_KEYS = {
     "ps_resource_request", "ps_resource_limit", "worker_pod_priority",
     "ps_pod_priority", "volume", "image_pull_policy", "restart_policy",
     "relaunch_on_worker_failure", "num_minibatches_per_shard", "job_type",
+    "envs", "data_reader_params", "cluster_spec",
 }
 
 

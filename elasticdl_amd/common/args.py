@@ -30,6 +30,8 @@ def add_common_train_params(p: argparse.ArgumentParser) -> None:
     p.add_argument("--training_data", default="")
     p.add_argument("--validation_data", default="")
     p.add_argument("--prediction_data", default="")
+    p.add_argument("--data_reader_params", default="",
+                   help="reader kwargs 'k=v;k2=v2' (e.g. CSV delimiter)")
     p.add_argument("--evaluation_steps", type=int, default=0)
     p.add_argument("--shuffle", type=str2bool, default=False)
     p.add_argument("--shuffle_shards", type=str2bool, default=False)
@@ -107,9 +109,30 @@ def parse_envs(s: str) -> dict:
     return out
 
 
+def _split_params(s: str):
+    """Split 'k=v;k2=v2' on ';' outside of quotes, so quoted values can
+    contain the separator: "delimiter=';'" is one pair."""
+    parts, buf, quote = [], [], ""
+    for ch in s or "":
+        if quote:
+            buf.append(ch)
+            if ch == quote:
+                quote = ""
+        elif ch in "'\"":
+            quote = ch
+            buf.append(ch)
+        elif ch == ";":
+            parts.append("".join(buf))
+            buf = []
+        else:
+            buf.append(ch)
+    parts.append("".join(buf))
+    return parts
+
+
 def parse_model_params(s: str) -> dict:
     out = {}
-    for part in (s or "").split(";"):
+    for part in _split_params(s):
         part = part.strip()
         if not part:
             continue

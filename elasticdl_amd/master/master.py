@@ -51,18 +51,24 @@ class Master:
         self.spec = spec
 
         # ---- data shards
+        reader_params = parse_model_params(
+            getattr(args, "data_reader_params", ""))
+
         def _make_reader(origin: str):
             """Same resolution chain for train/validation/prediction data:
             zoo custom_data_reader > generic synthetic:<n> > file factory."""
             if spec.data_reader_fn is not None:
-                return spec.data_reader_fn(origin)
+                try:
+                    return spec.data_reader_fn(origin, **reader_params)
+                except TypeError:
+                    return spec.data_reader_fn(origin)
             from elasticdl_amd.data.reader import (
                 create_data_reader,
                 synthetic_reader_from_spec,
             )
 
             return synthetic_reader_from_spec(spec, origin) \
-                or create_data_reader(origin)
+                or create_data_reader(origin, **reader_params)
 
         reader = None
         training_shards = evaluation_shards = None
@@ -164,6 +170,7 @@ class Master:
             "--training_data", a.training_data,
             "--validation_data", a.validation_data,
             "--prediction_data", a.prediction_data,
+            "--data_reader_params", a.data_reader_params,
             "--device", a.device,
             "--log_loss_steps", str(a.log_loss_steps),
         ]

@@ -31,18 +31,24 @@ def build_worker(args) -> Worker:
                           model_zoo=getattr(args, "model_zoo", ""))
     mc = MasterClient(master_addr, worker_id)
 
+    reader_params = parse_model_params(
+        getattr(args, "data_reader_params", ""))
+
     def _make_reader(origin: str):
         # same chain as the master: zoo custom_data_reader >
         # generic synthetic:<n> > file factory
         if spec.data_reader_fn is not None:
-            return spec.data_reader_fn(origin)
+            try:
+                return spec.data_reader_fn(origin, **reader_params)
+            except TypeError:
+                return spec.data_reader_fn(origin)
         from elasticdl_amd.data.reader import (
             create_data_reader,
             synthetic_reader_from_spec,
         )
 
         return synthetic_reader_from_spec(spec, origin) \
-            or create_data_reader(origin)
+            or create_data_reader(origin, **reader_params)
 
     reader = None
     eval_reader = None

@@ -45,3 +45,23 @@ def test_cluster_spec_hook(tmp_path):
     assert out.patched == "worker"
     # empty spec is a no-op
     assert ClusterSpec("").patch_pod(pod, "ps") is pod
+
+
+def test_master_command_forwards_envs_and_reader_params():
+    """CLI -> master pod command keeps --envs / --data_reader_params /
+    --cluster_spec (they configure the pods the MASTER creates, so
+    dropping them at the client boundary silently disables them in k8s
+    mode)."""
+    from elasticdl_amd.client.api import build_master_command
+    from elasticdl_amd.client.main import build_parser
+
+    args = build_parser().parse_args([
+        "train", "--model_def", "mnist", "--image_name", "img",
+        "--envs", "A=1,B=2",
+        "--data_reader_params", "delimiter=;",
+        "--cluster_spec", "my_spec.py",
+    ])
+    cmd = build_master_command(args)
+    assert cmd[cmd.index("--envs") + 1] == "A=1,B=2"
+    assert cmd[cmd.index("--data_reader_params") + 1] == "delimiter=;"
+    assert cmd[cmd.index("--cluster_spec") + 1] == "my_spec.py"

@@ -535,3 +535,35 @@ def test_hung_worker_killed_and_job_completes(tmp_path):
     out = r.stdout + r.stderr
     assert r.returncode == 0, out[-3000:]
     assert "considered hung" in out
+
+
+@pytest.mark.timeout(300)
+def test_data_reader_params_csv_delimiter(tmp_path):
+    """--data_reader_params reaches the reader factory (reference flag:
+    elasticdl_client/common/args.py:501): train iris from a
+    semicolon-delimited CSV."""
+    import random
+
+    rng = random.Random(0)
+    csv_path = tmp_path / "iris_semi.csv"
+    with open(csv_path, "w") as f:
+        f.write("f0;f1;f2;f3;label\n")
+        for _ in range(96):
+            c = rng.randint(0, 2)
+            f.write(";".join(f"{rng.gauss(c, 0.3):.3f}" for _ in range(4))
+                    + f";{c}\n")
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "iris",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--training_data", str(csv_path),
+        "--data_reader_params", "delimiter=';'",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
