@@ -52,6 +52,12 @@ def add_common_train_params(p: argparse.ArgumentParser) -> None:
     p.add_argument("--embedding_max_rows", type=int, default=1 << 22)
     p.add_argument("--device", default="auto")
     p.add_argument("--envs", default="", help="extra pod env 'k=v,k2=v2'")
+    p.add_argument("--populate_env_names", default="",
+                   help="regex of env var names the master copies from its "
+                        "own environment into every pod it creates")
+    p.add_argument("--log_level", default="",
+                   help="python logging level for this process tree "
+                        "(overrides EDL_LOG_LEVEL)")
 
 
 def add_k8s_params(p: argparse.ArgumentParser) -> None:
@@ -92,6 +98,19 @@ def parse_worker_args(argv=None) -> argparse.Namespace:
     p.add_argument("--worker_id", type=int, default=-1)
     p.add_argument("--ps_addrs", default="", help="comma-separated PS addresses")
     return p.parse_args(argv)
+
+
+def populated_envs(pattern: str) -> dict:
+    """Env vars whose NAME matches the regex, copied from this process's
+    environment (reference: pod_manager.py:153-158 populate_env_names) —
+    the way ROCm/MIOpen tuning vars reach worker/PS pods."""
+    if not pattern:
+        return {}
+    import os
+    import re
+
+    rx = re.compile(pattern)
+    return {k: v for k, v in os.environ.items() if rx.fullmatch(k)}
 
 
 def parse_envs(s: str) -> dict:

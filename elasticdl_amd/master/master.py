@@ -14,7 +14,11 @@ import threading
 import time
 from typing import List, Optional
 
-from elasticdl_amd.common.args import parse_envs, parse_model_params
+from elasticdl_amd.common.args import (
+    parse_envs,
+    parse_model_params,
+    populated_envs,
+)
 from elasticdl_amd.common.constants import DistributionStrategy
 from elasticdl_amd.common.log_utils import default_logger as logger
 from elasticdl_amd.common.rpc import start_server
@@ -46,6 +50,11 @@ class Master:
         self.stopped = threading.Event()
         self.exit_code = 0
 
+        if getattr(args, "log_level", ""):
+            import logging
+
+            logging.getLogger("elasticdl_amd").setLevel(
+                args.log_level.upper())
         spec = get_model_spec(args.model_def, parse_model_params(args.model_params),
                               model_zoo=getattr(args, "model_zoo", ""))
         self.spec = spec
@@ -224,7 +233,10 @@ class Master:
                 a, "relaunch_on_worker_failure", 3
             ),
             log_dir=os.path.join(a.checkpoint_dir or "/tmp/edl", "logs"),
-            user_envs=parse_envs(getattr(a, "envs", "")),
+            user_envs={
+                **populated_envs(getattr(a, "populate_env_names", "")),
+                **parse_envs(getattr(a, "envs", "")),
+            },
         )
         mgr.ps_addrs = self.ps_addrs
         mgr.add_pod_event_callback(TaskRescheduleCallback(self.task_manager))

@@ -126,9 +126,10 @@ class PodManager:
         owner = self.k8s.get_pod(self.k8s.get_master_pod_name())
         from elasticdl_amd.common.constants import WorkerEnv
 
-        from elasticdl_amd.common.args import parse_envs
+        from elasticdl_amd.common.args import parse_envs, populated_envs
 
-        envs = parse_envs(getattr(self.args, "envs", ""))
+        envs = populated_envs(getattr(self.args, "populate_env_names", ""))
+        envs.update(parse_envs(getattr(self.args, "envs", "")))
         envs.update({
             WorkerEnv.MASTER_ADDR: self.master.master_addr,
             WorkerEnv.WORKER_ID: str(index),

@@ -72,3 +72,42 @@ def test_k8s_pod_spec_carries_user_envs(monkeypatch):
     assert env["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
     # framework-owned vars are not overridable by --envs
     assert env["EDL_MASTER_ADDR"] == master.master_addr
+
+
+def test_populate_env_names_regex(monkeypatch):
+    from elasticdl_amd.common.args import populated_envs
+
+    monkeypatch.setenv("MIOPEN_FIND_MODE", "1")
+    monkeypatch.setenv("MIOPEN_USER_DB_PATH", "/x")
+    monkeypatch.setenv("UNRELATED", "z")
+    got = populated_envs(r"MIOPEN_.*")
+    assert got == {"MIOPEN_FIND_MODE": "1", "MIOPEN_USER_DB_PATH": "/x"}
+    assert populated_envs("") == {}
+
+
+def test_populate_env_names_reaches_k8s_pods(monkeypatch):
+    from tests.test_k8s_mode_e2e import FakeCluster
+
+    from elasticdl_amd.common.args import parse_master_args
+    from elasticdl_amd.master.k8s_client import Client
+    from elasticdl_amd.master.master import Master
+
+    fake = FakeCluster()
+    monkeypatch.setenv("MY_POD_IP", "10.0.0.8")
+    monkeypatch.setenv("HSA_XX_TEST_FLAG", "7")
+    args = parse_master_args([
+        "--pod_manager", "k8s",
+        "--job_name", "jobp",
+        "--image_name", "img",
+        "--model_def", "mnist",
+        "--num_workers", "1",
+        "--training_data", "synthetic:32",
+        "--populate_env_names", r"HSA_XX_.*",
+    ])
+    master = Master(args, k8s_client=Client("default", "jobp", "img",
+                                            core_api=fake))
+    master.pod_manager.start_workers()
+    pod = fake.pods["elasticdl-jobp-worker-0"]
+    env = {e.name: e.value for e in pod.spec.containers[0].env
+           if e.value is not None}
+    assert env["HSA_XX_TEST_FLAG"] == "7"
