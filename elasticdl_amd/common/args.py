@@ -77,6 +77,8 @@ def add_k8s_params(p: argparse.ArgumentParser) -> None:
     p.add_argument("--image_pull_policy", default="IfNotPresent")
     p.add_argument("--restart_policy", default="Never")
     p.add_argument("--cluster_spec", default="")
+    p.add_argument("--cluster_spec_json", default="",
+                   help="JSON dict of declarative pod/service additions")
     p.add_argument("--relaunch_on_worker_failure", type=int, default=3)
     p.add_argument("--yaml", default="", help="dump pod YAML instead of submitting")
 

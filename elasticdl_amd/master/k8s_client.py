@@ -75,26 +75,75 @@ def parse_volume(spec: str) -> List[Dict[str, str]]:
 
 
 class ClusterSpec:
-    """Bespoke-cluster pod mutation hook (reference:
-    elasticdl_client/common/k8s_client.py:106-219): a python module/file
-    exposing ``patch_pod(pod, pod_type) -> pod`` applied to every pod spec
-    before creation."""
+    """Bespoke-cluster pod/service mutation hook (reference:
+    elasticdl_client/common/k8s_client.py:106-219). Two forms, composable:
 
-    def __init__(self, spec: str = ""):
+    - ``spec``: a python module/file exposing ``patch_pod(pod, pod_type)``
+      (and optionally ``patch_service(service)``);
+    - ``spec_json``: a JSON dict with declarative additions —
+      ``pod_spec`` (all pods), ``master_spec``/``ps_spec``/``worker_spec``
+      (per type), ``service_spec``; each may set labels, annotations,
+      env (name/value pairs), tolerations, affinity, node_selector.
+    """
+
+    def __init__(self, spec: str = "", spec_json: str = ""):
         self._mod = None
+        self._json = None
         if spec:
             from elasticdl_amd.utils.model_utils import load_module
 
             self._mod = load_module(spec)
+        if spec_json:
+            import json
+
+            self._json = json.loads(spec_json)
+
+    @staticmethod
+    def _apply_pod_spec(pod, d: dict):
+        meta = pod.metadata
+        if "labels" in d:
+            meta.labels = {**(meta.labels or {}), **d["labels"]}
+        if "annotations" in d:
+            meta.annotations = {
+                **(getattr(meta, "annotations", None) or {}),
+                **d["annotations"],
+            }
+        k8s = k8s_types()
+        for e in d.get("env", []):
+            for c in pod.spec.containers:
+                c.env = list(c.env or [])
+                c.env.append(k8s.V1EnvVar(name=e["name"], value=e["value"]))
+        if "tolerations" in d:
+            pod.spec.tolerations = d["tolerations"]
+        if "affinity" in d:
+            pod.spec.affinity = d["affinity"]
+        if "node_selector" in d:
+            pod.spec.node_selector = d["node_selector"]
+        return pod
 
     def patch_pod(self, pod, pod_type: str):
         if self._mod is not None and hasattr(self._mod, "patch_pod"):
-            return self._mod.patch_pod(pod, pod_type) or pod
+            pod = self._mod.patch_pod(pod, pod_type) or pod
+        if self._json is not None:
+            for key in ("pod_spec", f"{pod_type}_spec"):
+                if key in self._json:
+                    pod = self._apply_pod_spec(pod, self._json[key])
         return pod
 
     def patch_service(self, service):
         if self._mod is not None and hasattr(self._mod, "patch_service"):
-            return self._mod.patch_service(service) or service
+            service = self._mod.patch_service(service) or service
+        if self._json is not None and "service_spec" in self._json:
+            d = self._json["service_spec"]
+            meta = service.metadata
+            if "labels" in d:
+                meta.labels = {**(getattr(meta, "labels", None) or {}),
+                               **d["labels"]}
+            if "annotations" in d:
+                meta.annotations = {
+                    **(getattr(meta, "annotations", None) or {}),
+                    **d["annotations"],
+                }
         return service
 
 

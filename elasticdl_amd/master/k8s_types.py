@@ -73,7 +73,8 @@ class V1Container(_Spec):
 
 class V1PodSpec(_Spec):
     _fields = ("containers", "restart_policy", "priority_class_name",
-               "volumes", "termination_grace_period_seconds")
+               "volumes", "termination_grace_period_seconds",
+               "tolerations", "affinity", "node_selector")
 
 
 class V1OwnerReference(_Spec):

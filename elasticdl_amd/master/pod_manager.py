@@ -54,7 +54,10 @@ class PodManager:
         )
         from elasticdl_amd.master.k8s_client import ClusterSpec
 
-        self.cluster_spec = ClusterSpec(getattr(args, "cluster_spec", ""))
+        self.cluster_spec = ClusterSpec(
+            getattr(args, "cluster_spec", ""),
+            getattr(args, "cluster_spec_json", ""),
+        )
         self._lock = threading.Lock()
         self.pods: Dict[str, PodInfo] = {}
         self._next_worker_id = 0

@@ -65,3 +65,58 @@ def test_master_command_forwards_envs_and_reader_params():
     assert cmd[cmd.index("--envs") + 1] == "A=1,B=2"
     assert cmd[cmd.index("--data_reader_params") + 1] == "delimiter=;"
     assert cmd[cmd.index("--cluster_spec") + 1] == "my_spec.py"
+
+
+def test_cluster_spec_json_patches_pods_and_services(monkeypatch):
+    """Declarative JSON cluster spec (reference ClusterSpec json form):
+    pod_spec applies to all pods, worker_spec only to workers,
+    service_spec to services."""
+    import json
+
+    from tests.test_k8s_mode_e2e import FakeCluster
+
+    from elasticdl_amd.common.args import parse_master_args
+    from elasticdl_amd.master.k8s_client import Client
+    from elasticdl_amd.master.master import Master
+
+    spec = {
+        "pod_spec": {
+            "labels": {"team": "ctr"},
+            "env": [{"name": "RCCL_DEBUG", "value": "WARN"}],
+            "tolerations": [{"key": "amd.com/gpu", "operator": "Exists"}],
+        },
+        "worker_spec": {"annotations": {"role": "trainer"}},
+        "service_spec": {"labels": {"svc": "ps"}},
+    }
+    fake = FakeCluster()
+    monkeypatch.setenv("MY_POD_IP", "10.0.0.7")
+    args = parse_master_args([
+        "--pod_manager", "k8s",
+        "--job_name", "jobj",
+        "--image_name", "img",
+        "--model_def", "mnist",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--training_data", "synthetic:32",
+        "--cluster_spec_json", json.dumps(spec),
+    ])
+    master = Master(args, k8s_client=Client("default", "jobj", "img",
+                                            core_api=fake))
+    master.pod_manager.start_parameter_servers()
+    master.pod_manager.start_workers()
+
+    w = fake.pods["elasticdl-jobj-worker-0"]
+    assert w.metadata.labels["team"] == "ctr"
+    assert w.metadata.annotations["role"] == "trainer"
+    env = {e.name: e.value for e in w.spec.containers[0].env
+           if e.value is not None}
+    assert env["RCCL_DEBUG"] == "WARN"
+    assert w.spec.tolerations[0]["key"] == "amd.com/gpu"
+
+    ps = fake.pods["elasticdl-jobj-ps-0"]
+    assert ps.metadata.labels["team"] == "ctr"
+    assert getattr(ps.metadata, "annotations", None) in (None, {}) or \
+        "role" not in ps.metadata.annotations
+
+    svc = fake.services["elasticdl-jobj-ps-0"]
+    assert svc.metadata.labels["svc"] == "ps"
