@@ -183,6 +183,8 @@ class Master:
             "--device", a.device,
             "--log_loss_steps", str(a.log_loss_steps),
         ]
+        if getattr(a, "log_level", ""):
+            cmd += ["--log_level", a.log_level]
         if a.output:
             cmd += ["--output", a.output]
         if self.ps_addrs:

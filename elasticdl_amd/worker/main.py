@@ -27,6 +27,10 @@ def build_worker(args) -> Worker:
     if device == "auto":
         device = "cuda" if torch.cuda.is_available() else "cpu"
 
+    if getattr(args, "log_level", ""):
+        import logging
+
+        logging.getLogger("elasticdl_amd").setLevel(args.log_level.upper())
     spec = get_model_spec(args.model_def, parse_model_params(args.model_params),
                           model_zoo=getattr(args, "model_zoo", ""))
     mc = MasterClient(master_addr, worker_id)
