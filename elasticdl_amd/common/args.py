@@ -55,6 +55,11 @@ def add_common_train_params(p: argparse.ArgumentParser) -> None:
     p.add_argument("--populate_env_names", default="",
                    help="regex of env var names the master copies from its "
                         "own environment into every pod it creates")
+    p.add_argument("--job_command", default="",
+                   help="custom command run in worker pods instead of the "
+                        "built-in worker runtime (SDK-style jobs: the "
+                        "command reads EDL_MASTER_ADDR / EDL_WORKER_ID and "
+                        "drives its own elastic loop)")
     p.add_argument("--log_level", default="",
                    help="python logging level for this process tree "
                         "(overrides EDL_LOG_LEVEL)")

@@ -166,6 +166,11 @@ class Master:
     # ------------------------------------------------------------ commands
     def worker_command(self, worker_id: int) -> List[str]:
         a = self.args
+        if getattr(a, "job_command", ""):
+            # SDK-style job (reference pod_manager.py:327-380): the worker
+            # pod runs the user's own command; coordinates come from the
+            # EDL_* env the pod managers inject.
+            return ["bash", "-c", a.job_command]
         cmd = [
             sys.executable, "-m", "elasticdl_amd.worker.main",
             "--master_addr", self.master_addr,

@@ -567,3 +567,49 @@ def test_data_reader_params_csv_delimiter(tmp_path):
     r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
                        cwd=REPO, capture_output=True, text=True, timeout=280)
     assert r.returncode == 0, r.stderr[-3000:]
+
+
+@pytest.mark.timeout(300)
+def test_job_command_sdk_worker(tmp_path):
+    """--job_command (reference pod_manager.py:327-380): worker pods run
+    the user's own program, which coordinates through EDL_* env +
+    MasterClient/DataShardService — the SDK-job launch path."""
+    marker = str(tmp_path / "done.txt")
+    script = tmp_path / "sdk_worker.py"
+    script.write_text(textwrap.dedent(f"""
+        import os, time
+        from elasticdl_amd.worker.master_client import MasterClient
+        from elasticdl_amd.worker.data_shard_service import DataShardService
+        from elasticdl_amd.common.task import TaskType
+
+        mc = MasterClient(os.environ["EDL_MASTER_ADDR"],
+                          int(os.environ["EDL_WORKER_ID"]))
+        svc = DataShardService(mc, 16)
+        n = 0
+        while True:
+            t = svc.fetch_task()
+            if t.type == TaskType.NONE:
+                break
+            if t.type == TaskType.WAIT:
+                time.sleep(0.2)
+                continue
+            n += t.shard.size
+            svc.report_batch_done(t.shard.size)
+        open({marker!r}, "w").write(str(n))
+    """))
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "mnist",
+        "--num_workers", "1",
+        "--num_minibatches_per_task", "2",
+        "--minibatch_size", "16",
+        "--training_data", "synthetic:96",
+        "--job_command", f"{sys.executable} {script}",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert os.path.exists(marker)
+    assert open(marker).read() == "96"
