@@ -27,7 +27,8 @@ _MASTER_ARG_KEYS = {
     "lr_staleness_modulation", "sync_version_tolerance", "get_model_steps",
     "checkpoint_dir", "checkpoint_steps", "keep_checkpoint_max",
     "checkpoint_dir_for_init", "output", "log_loss_steps",
-    "task_timeout_sec", "embedding_max_rows", "device", "namespace",
+    "task_timeout_sec", "task_fault_tolerance", "relaunch_timeout_worker",
+    "embedding_max_rows", "device", "namespace",
     "image_name", "worker_resource_request", "worker_resource_limit",
     "ps_resource_request", "ps_resource_limit", "worker_pod_priority",
     "ps_pod_priority", "volume", "image_pull_policy", "restart_policy",

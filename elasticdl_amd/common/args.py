@@ -49,6 +49,8 @@ def add_common_train_params(p: argparse.ArgumentParser) -> None:
     p.add_argument("--output", default="", help="model export path")
     p.add_argument("--log_loss_steps", type=int, default=100)
     p.add_argument("--task_timeout_sec", type=float, default=300.0)
+    p.add_argument("--task_fault_tolerance", type=str2bool, default=True)
+    p.add_argument("--relaunch_timeout_worker", type=str2bool, default=True)
     p.add_argument("--embedding_max_rows", type=int, default=1 << 22)
     p.add_argument("--device", default="auto")
     p.add_argument("--envs", default="", help="extra pod env 'k=v,k2=v2'")

@@ -111,6 +111,9 @@ class Master:
             shuffle=args.shuffle,
             shuffle_shards=args.shuffle_shards,
             task_timeout_sec=args.task_timeout_sec,
+            task_fault_tolerance=getattr(args, "task_fault_tolerance", True),
+            relaunch_timeout_worker=getattr(
+                args, "relaunch_timeout_worker", True),
         )
         if self.job_type == "evaluate":
             self.task_manager.create_evaluation_tasks(model_version=0)

@@ -53,6 +53,8 @@ class TaskManager:
         shuffle: bool = False,
         shuffle_shards: bool = False,
         task_timeout_sec: float = _MIN_TASK_TIMEOUT_SEC,
+        task_fault_tolerance: bool = True,
+        relaunch_timeout_worker: bool = True,
     ):
         self._lock = threading.Lock()
         self._training_shards = list(training_shards or [])
@@ -66,6 +68,10 @@ class TaskManager:
         self._shuffle = shuffle
         self._shuffle_shards = shuffle_shards
         self._task_timeout_sec = max(task_timeout_sec, 1e-3)
+        # reference task_manager.py:126-127 + :390/:467/:546/:570 — strict
+        # mode: no task recovery/requeue and no hung-worker watchdog
+        self._fault_tolerance = task_fault_tolerance
+        self._relaunch_timeout_worker = relaunch_timeout_worker
 
         self._todo: List[Task] = []
         self._doing: Dict[int, _DoingEntry] = {}
@@ -301,7 +307,7 @@ class TaskManager:
             else:
                 retries = self._task_retry_count.get(task_id, 0) + 1
                 self._task_retry_count[task_id] = retries
-                if retries <= MAX_TASK_RETRIES:
+                if self._fault_tolerance and retries <= MAX_TASK_RETRIES:
                     logger.info(
                         "Task %d failed (retry %d/%d); requeueing",
                         task_id,
@@ -320,6 +326,8 @@ class TaskManager:
     def recover_tasks(self, worker_id: int) -> int:
         """Requeue all doing tasks of a dead worker
         (reference: task_manager.py:544-560)."""
+        if not self._fault_tolerance:
+            return 0
         with self._lock:
             ids = [
                 tid
@@ -337,6 +345,8 @@ class TaskManager:
 
     # --------------------------------------------------------------- watchdog
     def start(self) -> None:
+        if not (self._fault_tolerance and self._relaunch_timeout_worker):
+            return
         if self._watchdog is None:
             self._watchdog = threading.Thread(
                 target=self._watchdog_loop, name="task-watchdog", daemon=True

@@ -166,3 +166,24 @@ def test_worker_driven_params():
     )
     c = tm.counts()
     assert c["todo"] == 2  # 100 / (10*5)
+
+
+def test_strict_mode_disables_recovery_and_requeue():
+    """--task_fault_tolerance false (reference task_manager.py:126,
+    390, 467, 546): no dead-worker recovery, failed tasks drop on first
+    failure, no watchdog thread."""
+    from elasticdl_amd.common.task import TaskType
+
+    tm = TaskManager(training_shards=[("s", 0, 64)], records_per_task=32,
+                     task_fault_tolerance=False)
+    tm.start()
+    assert tm._watchdog is None
+    t1 = tm.get(0)
+    assert t1.type == TaskType.TRAINING
+    assert tm.recover_tasks(0) == 0          # strict: no recovery
+    tm.report(t1.task_id, False, 0)          # strict: dropped, not requeued
+    t2 = tm.get(1)
+    assert t2.task_id != t1.task_id
+    tm.report(t2.task_id, True, 1)
+    assert tm.get(1).type == TaskType.NONE
+    assert tm.failed_records == 32
