@@ -34,6 +34,8 @@ _MASTER_ARG_KEYS = {
     "ps_pod_priority", "volume", "image_pull_policy", "restart_policy",
     "relaunch_on_worker_failure", "num_minibatches_per_shard", "job_type",
     "envs", "data_reader_params", "cluster_spec",
+    "loss", "optimizer", "feed", "eval_metrics_fn", "callbacks",
+    "custom_data_reader",
     "populate_env_names", "log_level", "cluster_spec_json", "job_command",
 }
 

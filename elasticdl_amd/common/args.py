@@ -20,6 +20,10 @@ def add_common_train_params(p: argparse.ArgumentParser) -> None:
                    help="zoo module (file, dotted path, or builtin name)")
     p.add_argument("--model_params", default="",
                    help="kwargs for custom_model(), 'k=v;k2=v2'")
+    for fn_flag in ("loss", "optimizer", "feed", "eval_metrics_fn",
+                    "callbacks", "custom_data_reader"):
+        p.add_argument(f"--{fn_flag}", default="",
+                       help=f"custom {fn_flag} function name in the module")
     p.add_argument("--distribution_strategy", default="Local",
                    choices=["Local", "ParameterServerStrategy",
                             "AllreduceStrategy"])
@@ -188,3 +192,17 @@ def build_arguments_from_parsed_result(args: argparse.Namespace,
             continue
         out.extend([f"--{k}", str(v)])
     return out
+
+
+def function_names_from_args(args) -> dict:
+    """Collect the --loss/--optimizer/... function-name overrides."""
+    return {
+        k: v for k, v in (
+            ("loss", getattr(args, "loss", "")),
+            ("optimizer", getattr(args, "optimizer", "")),
+            ("feed", getattr(args, "feed", "")),
+            ("eval_metrics_fn", getattr(args, "eval_metrics_fn", "")),
+            ("callbacks", getattr(args, "callbacks", "")),
+            ("custom_data_reader", getattr(args, "custom_data_reader", "")),
+        ) if v
+    }

@@ -15,6 +15,7 @@ import time
 from typing import List, Optional
 
 from elasticdl_amd.common.args import (
+    function_names_from_args,
     parse_envs,
     parse_model_params,
     populated_envs,
@@ -56,7 +57,8 @@ class Master:
             logging.getLogger("elasticdl_amd").setLevel(
                 args.log_level.upper())
         spec = get_model_spec(args.model_def, parse_model_params(args.model_params),
-                              model_zoo=getattr(args, "model_zoo", ""))
+                              model_zoo=getattr(args, "model_zoo", ""),
+                              function_names=function_names_from_args(args))
         self.spec = spec
 
         # ---- data shards
@@ -193,6 +195,11 @@ class Master:
         ]
         if getattr(a, "log_level", ""):
             cmd += ["--log_level", a.log_level]
+        for fn_flag in ("loss", "optimizer", "feed", "eval_metrics_fn",
+                        "callbacks", "custom_data_reader"):
+            v = getattr(a, fn_flag, "")
+            if v:
+                cmd += [f"--{fn_flag}", v]
         if a.output:
             cmd += ["--output", a.output]
         if self.ps_addrs:

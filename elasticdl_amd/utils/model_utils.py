@@ -81,12 +81,18 @@ class ModelSpec:
 
 
 def get_model_spec(model_def: str, model_params: Optional[dict] = None,
-                   model_zoo: str = "") -> ModelSpec:
+                   model_zoo: str = "",
+                   function_names: Optional[dict] = None) -> ModelSpec:
+    """``function_names`` maps spec slots to custom function names in the
+    module (reference flags --loss/--optimizer/--feed/... default to the
+    conventional names), e.g. {"loss": "my_loss"}."""
     mod = load_module(model_def, model_zoo)
+    names = dict(function_names or {})
 
-    def need(name):
+    def pick(slot, default, required):
+        name = names.get(slot) or default
         fn = getattr(mod, name, None)
-        if fn is None:
+        if fn is None and required:
             raise AttributeError(
                 f"model zoo module {model_def!r} must define {name}()"
             )
@@ -94,13 +100,14 @@ def get_model_spec(model_def: str, model_params: Optional[dict] = None,
 
     return ModelSpec(
         module=mod,
-        model_fn=need("custom_model"),
-        loss_fn=need("loss"),
-        optimizer_fn=need("optimizer"),
-        feed_fn=getattr(mod, "feed", None),
-        eval_metrics_fn=getattr(mod, "eval_metrics_fn", None),
-        data_reader_fn=getattr(mod, "custom_data_reader", None),
-        callbacks_fn=getattr(mod, "callbacks", None),
+        model_fn=pick("model", "custom_model", True),
+        loss_fn=pick("loss", "loss", True),
+        optimizer_fn=pick("optimizer", "optimizer", True),
+        feed_fn=pick("feed", "feed", False),
+        eval_metrics_fn=pick("eval_metrics_fn", "eval_metrics_fn", False),
+        data_reader_fn=pick("custom_data_reader", "custom_data_reader",
+                            False),
+        callbacks_fn=pick("callbacks", "callbacks", False),
         params=model_params or {},
     )
 

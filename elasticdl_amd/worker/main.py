@@ -8,7 +8,7 @@ import sys
 
 import torch
 
-from elasticdl_amd.common.args import parse_model_params, parse_worker_args
+from elasticdl_amd.common.args import function_names_from_args, parse_model_params, parse_worker_args
 from elasticdl_amd.common.constants import DistributionStrategy, WorkerEnv
 from elasticdl_amd.common.log_utils import default_logger as logger
 from elasticdl_amd.utils.model_utils import get_model_spec
@@ -32,7 +32,8 @@ def build_worker(args) -> Worker:
 
         logging.getLogger("elasticdl_amd").setLevel(args.log_level.upper())
     spec = get_model_spec(args.model_def, parse_model_params(args.model_params),
-                          model_zoo=getattr(args, "model_zoo", ""))
+                          model_zoo=getattr(args, "model_zoo", ""),
+                          function_names=function_names_from_args(args))
     mc = MasterClient(master_addr, worker_id)
 
     reader_params = parse_model_params(

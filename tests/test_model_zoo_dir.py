@@ -613,3 +613,32 @@ def test_job_command_sdk_worker(tmp_path):
     assert r.returncode == 0, r.stderr[-3000:]
     assert os.path.exists(marker)
     assert open(marker).read() == "96"
+
+
+@pytest.mark.timeout(300)
+def test_function_name_selector_flags(zoo, tmp_path):
+    """--loss/--optimizer/... select custom function names in the module
+    (reference flags, elasticdl_client/common/args.py): a module whose
+    loss is called my_loss trains when --loss my_loss is passed."""
+    (tmp_path / "z").mkdir()
+    (tmp_path / "z" / "named.py").write_text(
+        ZOO_MODULE.replace("def loss(", "def my_loss(")
+                  .replace("def optimizer(", "def build_opt(")
+    )
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_zoo", str(tmp_path / "z"),
+        "--model_def", "named",
+        "--loss", "my_loss",
+        "--optimizer", "build_opt",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "16",
+        "--num_minibatches_per_task", "2",
+        "--training_data", "synthetic:64",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=REPO),
+                       cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
