@@ -66,3 +66,49 @@ def test_factory(tmp_path):
     # resolves to the offline-testable stub (see test_recordio.py)
     with pytest.raises(ValueError):
         create_data_reader("odps://project/table")
+
+
+def test_odps_reader_full_job_with_injected_client():
+    """ODPS read path end to end (in-process): table-size sharding,
+    range reads through the injected client, records collated and
+    trained by the real Worker/LocalTrainer."""
+    import torch
+
+    from elasticdl_amd.common import rpc
+    from elasticdl_amd.data.reader import ODPSReader
+    from elasticdl_amd.master.servicer import MasterServicer
+    from elasticdl_amd.master.task_manager import TaskManager
+    from elasticdl_amd.utils.model_utils import get_model_spec
+    from elasticdl_amd.worker.master_client import MasterClient
+    from elasticdl_amd.worker.trainer import LocalTrainer
+    from elasticdl_amd.worker.worker import Worker
+
+    class FakeODPS:
+        def table_size(self, project, table):
+            assert (project, table) == ("proj", "iris_table")
+            return 64
+
+        def read_rows(self, project, table, start, end):
+            for i in range(start, end):
+                g = torch.Generator().manual_seed(i)
+                c = i % 3
+                x = (torch.randn(4, generator=g) * 0.2 + c).tolist()
+                yield [*x, c]
+
+    spec = get_model_spec("iris")
+    reader = ODPSReader("odps://proj/tables/iris_table",
+                        records_per_shard=16, client=FakeODPS())
+    shards = reader.create_shards()
+    assert shards == [("proj/iris_table", lo, lo + 16)
+                      for lo in range(0, 64, 16)]
+    tm = TaskManager(training_shards=shards, records_per_task=16)
+    server = rpc.start_server(
+        "127.0.0.1:0", {"Master": MasterServicer(tm).methods()})
+    try:
+        mc = MasterClient(f"127.0.0.1:{server.port}", worker_id=0)
+        Worker(0, mc, LocalTrainer(spec, device="cpu"), data_reader=reader,
+               spec=spec, minibatch_size=16).run()
+        assert tm.finished()
+        assert tm.counts()["completed_records"] == 64
+    finally:
+        server.stop(0)
