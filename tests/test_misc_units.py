@@ -141,3 +141,45 @@ def test_task_manager_throughput_counters():
     c = tm.counts()
     assert c["completed_records"] == 64
     assert c["records_per_sec"] >= 0
+
+
+def test_timing_sections_and_worker_timing_log(tmp_path):
+    """EDL_TIMING observability (reference common/timing_utils.py:17-48 +
+    per-task report, worker.py:374-376): sections accumulate and the
+    worker logs a per-task timing line when enabled."""
+    import os
+    import subprocess
+    import sys
+
+    from elasticdl_amd.common.timing import Timing
+
+    t = Timing(enabled=True)
+    t.start_record_time("a")
+    t.end_record_time("a")
+    t.start_record_time("a")
+    t.end_record_time("a")
+    t.start_record_time("b")
+    t.end_record_time("b")
+    rep = t.report_timing(reset=True)
+    assert "a:" in rep and "b:" in rep and "ms" in rep
+    assert t.report_timing() == ""
+    # disabled -> no-ops
+    t2 = Timing(enabled=False)
+    t2.start_record_time("x")
+    t2.end_record_time("x")
+    assert t2.report_timing() == ""
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "elasticdl_amd.master.main",
+         "--model_def", "mnist", "--num_workers", "1",
+         "--training_data", "synthetic:32", "--minibatch_size", "16",
+         "--num_minibatches_per_task", "2", "--device", "cpu",
+         "--checkpoint_dir", str(tmp_path),
+         "--pod_manager", "local"],
+        env=dict(os.environ, PYTHONPATH=repo, EDL_TIMING="1"),
+        cwd=repo, capture_output=True, text=True, timeout=280,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    wlog = open(os.path.join(str(tmp_path), "logs", "worker-0.log")).read()
+    assert "timing:" in wlog and "batch_process" in wlog
