@@ -79,6 +79,15 @@ class MasterServicer:
             )
             if len(alive) > 1 and worker_id != alive[0]:
                 task = Task(task_id=0, shard=None, type=TaskType.NONE)
+                # stage its rendezvous removal NOW: survivors would
+                # otherwise block a full store timeout on a world that
+                # still lists the exiting worker (it also reports END on
+                # exit, but this closes the race window)
+                try:
+                    self._rendezvous_server.remove_worker(
+                        f"worker-{worker_id}")
+                except Exception:  # noqa: BLE001 - best-effort
+                    pass
         return task.to_wire()
 
     def report_task_result(self, req: dict) -> dict:
