@@ -23,6 +23,7 @@ _BUILTIN = {
     "cifar10": "elasticdl_amd.models.cifar10",
     "iris": "elasticdl_amd.models.iris",
     "census_wide_deep": "elasticdl_amd.models.census_wide_deep",
+    "census_dnn": "elasticdl_amd.models.census_dnn",
     "mobilenetv2": "elasticdl_amd.models.mobilenetv2",
     "heart": "elasticdl_amd.models.heart",
 }

@@ -237,3 +237,66 @@ def test_heart_model_trains():
     csv_row = ["63", "145", "233", "150", "2.3", "3", "0", "fixed", "1"]
     f2, l2 = zoo.collate_fn([csv_row])
     assert f2["thal"] == ["fixed"] and int(l2[0]) == 1
+
+
+def test_census_dnn_trains_locally():
+    """census_dnn (reference model_zoo/census_dnn_model): the deep-only
+    census model learns on the same feature pipeline."""
+    from elasticdl_amd.models import census_dnn as zoo
+
+    model = zoo.custom_model()
+    eng = PSEngine(opt_type="adam", opt_args="learning_rate=0.01")
+    bind_local_engine(model, eng)
+    sink = []
+    for e in find_edl_embeddings(model):
+        e.set_grad_sink(sink)
+
+    reader = zoo.custom_data_reader("synthetic:64")
+    shards = reader.create_shards()
+    from elasticdl_amd.common.task import Shard, Task, TaskType
+
+    task = Task(1, Shard(*shards[0]), TaskType.TRAINING)
+    rows = list(reader.read_records(task))
+    feats, labels = zoo.feed(zoo.collate_fn(rows[:32]), "cpu")
+
+    opt = torch.optim.Adam(model.parameters(), lr=0.01)
+    losses = []
+    for _ in range(15):
+        sink.clear()
+        opt.zero_grad()
+        loss = zoo.loss(model(feats), labels)
+        loss.backward()
+        opt.step()
+        from elasticdl_amd.common.tensor_utils import merge_indexed_slices
+
+        edl = {}
+        for name, s in sink:
+            edl.setdefault(name, []).append(s)
+        eng.push_gradients(
+            {}, {n: merge_indexed_slices(*lst) for n, lst in edl.items()}
+        )
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses[::5]
+
+
+def test_census_dnn_end_to_end_local_job(tmp_path):
+    """Full subprocess job for the census_dnn zoo entry."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "elasticdl_amd.master.main",
+        "--model_def", "census_dnn",
+        "--distribution_strategy", "ParameterServerStrategy",
+        "--num_workers", "1", "--num_ps_pods", "1",
+        "--minibatch_size", "32",
+        "--num_minibatches_per_task", "2",
+        "--training_data", "synthetic:128",
+        "--device", "cpu",
+        "--pod_manager", "local",
+    ]
+    r = subprocess.run(cmd, env=dict(os.environ, PYTHONPATH=repo),
+                       cwd=repo, capture_output=True, text=True, timeout=280)
+    assert r.returncode == 0, r.stderr[-3000:]
