@@ -244,6 +244,7 @@ def test_census_dnn_trains_locally():
     census model learns on the same feature pipeline."""
     from elasticdl_amd.models import census_dnn as zoo
 
+    torch.manual_seed(0)  # learning assertion over few steps: fix init
     model = zoo.custom_model()
     eng = PSEngine(opt_type="adam", opt_args="learning_rate=0.01")
     bind_local_engine(model, eng)
