@@ -101,6 +101,7 @@ def test_to_ragged_to_sparse_layers():
 def test_census_wide_deep_trains_locally():
     from elasticdl_amd.models import census_wide_deep as zoo
 
+    torch.manual_seed(0)
     model = zoo.custom_model()
     eng = PSEngine(opt_type="adam", opt_args="learning_rate=0.01")
     bind_local_engine(model, eng)
@@ -214,6 +215,7 @@ def test_census_end_to_end_local_job(tmp_path):
 def test_heart_model_trains():
     from elasticdl_amd.models import heart as zoo
 
+    torch.manual_seed(0)
     model = zoo.custom_model()
     eng = PSEngine(opt_type="adam", opt_args="learning_rate=0.02")
     bind_local_engine(model, eng)

@@ -315,6 +315,8 @@ def test_get_model_steps_local_updates():
     spec = get_model_spec("iris")
     import torch
 
+    torch.manual_seed(0)
+
     def sample(i):
         g = torch.Generator().manual_seed(i)
         c = i % 3
