@@ -329,3 +329,21 @@ def synthetic_reader_from_spec(spec, data_origin: str,
 
     return SyntheticReader(size, sample, records_per_shard,
                            name=f"{spec.module.__name__}-synthetic")
+
+
+def call_data_reader_fn(fn, origin: str, params: dict):
+    """Call a zoo custom_data_reader with --data_reader_params kwargs when
+    its signature accepts them — signature-checked rather than
+    try/except TypeError, which would mask real errors inside the fn."""
+    if not params:
+        return fn(origin)
+    import inspect
+
+    try:
+        sig = inspect.signature(fn)
+        accepts = any(
+            p.kind == p.VAR_KEYWORD for p in sig.parameters.values()
+        ) or all(k in sig.parameters for k in params)
+    except (TypeError, ValueError):
+        accepts = False
+    return fn(origin, **params) if accepts else fn(origin)

@@ -69,10 +69,10 @@ class Master:
             """Same resolution chain for train/validation/prediction data:
             zoo custom_data_reader > generic synthetic:<n> > file factory."""
             if spec.data_reader_fn is not None:
-                try:
-                    return spec.data_reader_fn(origin, **reader_params)
-                except TypeError:
-                    return spec.data_reader_fn(origin)
+                from elasticdl_amd.data.reader import call_data_reader_fn
+
+                return call_data_reader_fn(
+                    spec.data_reader_fn, origin, reader_params)
             from elasticdl_amd.data.reader import (
                 create_data_reader,
                 synthetic_reader_from_spec,
