@@ -95,11 +95,15 @@ def test_allreduce_train_end_export(tmp_path):
     """--output in AllReduce mode: the train-end callback task routes to
     one worker which exports the final model."""
     export = str(tmp_path / "model.pt")
+    # EDL_MIN_WORLD: co-start barrier removes the worker-exits-during-
+    # peer-init race, which under heavy parallel-CI load can spiral into
+    # repeated 20 s store timeouts (serial runs don't need it)
     p = run_master([
         "--num_workers", "2",
-        "--training_data", "synthetic:128",
+        "--training_data", "synthetic:256",
         "--output", export,
-    ])
+    ], env_extra={"EDL_MIN_WORLD": "2",
+                  "EDL_MIN_WORLD_TIMEOUT_SEC": "60"})
     out, _ = p.communicate(timeout=600)
     assert p.returncode == 0, out[-3000:]
     assert os.path.exists(export), out[-2000:]
